@@ -1,0 +1,148 @@
+"""Contamination-threshold computation (distributed-ready).
+
+Mirrors SharedTrainLogic.computeAndSetModelThreshold
+(SharedTrainLogic.scala:175-242):
+
+* contamination == 0.0  => no threshold is set (model emits label 0.0).
+* contamination  > 0.0  => threshold = the (1 - contamination) quantile of
+  the TRAINING scores; ``contaminationError == 0.0`` demands the exact
+  order statistic (the reference's approxQuantile with relativeError 0,
+  SharedTrainLogic.scala:191-198), otherwise an approximate quantile
+  within that rank error is allowed.
+* afterwards the observed contamination (fraction of scores >= threshold)
+  is verified within ±verificationError (== contaminationError, or 1% of
+  contamination when exact) and a warning is logged on mismatch
+  (SharedTrainLogic.scala:211-232).
+
+The multi-GPU path replaces Spark's Greenwald-Khanna sketch merge with
+iterative histogram refinement: each pass is one 4096-bin histogram of the
+local shard + an all-reduce, narrowing the bracket around the target rank
+until the rank error bound (or float32 resolution) is met. The exact mode
+runs the same loop to float32 ULP resolution, then an exact distributed
+count-and-min pass picks the true order statistic.
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Optional
+
+import numpy as np
+import torch
+
+logger = logging.getLogger(__name__)
+
+_NBINS = 4096
+
+
+def _allreduce(t: torch.Tensor, comm) -> torch.Tensor:
+    if comm is not None:
+        comm.all_reduce(t)
+    return t
+
+
+def compute_threshold(
+    scores: torch.Tensor,
+    contamination: float,
+    contamination_error: float,
+    comm=None,
+) -> Optional[float]:
+    """Quantile threshold over (globally sharded) scores; None if skipped."""
+    if contamination <= 0.0:
+        return None
+    scores = scores.detach()
+    if scores.dtype not in (torch.float32, torch.float64):
+        scores = scores.float()
+
+    n_local = torch.tensor([scores.numel()], dtype=torch.float64, device=scores.device)
+    n_total = int(_allreduce(n_local.clone(), comm).item())
+    if n_total == 0:
+        raise ValueError("cannot compute threshold over zero scores")
+
+    # target: the k-th smallest with k = ceil((1-contamination) * n), 1-based
+    phi = 1.0 - contamination
+    k = max(1, min(n_total, int(np.ceil(phi * n_total))))
+
+    exact = contamination_error == 0.0
+    # allowed rank slack (in elements)
+    slack = 0 if exact else max(0, int(np.floor(contamination_error * n_total)))
+
+    if comm is None and exact:
+        # single-process exact: kthvalue (GPU radix select / CPU partition)
+        return float(torch.kthvalue(scores.flatten().float(), k).values.item())
+
+    if comm is not None:
+        gmin = comm.all_reduce_min(float(scores.min().item()))
+        gmax = comm.all_reduce_max(float(scores.max().item()))
+    else:
+        gmin = float(scores.min().item())
+        gmax = float(scores.max().item())
+
+    if gmin == gmax:
+        return gmin
+
+    s = scores.flatten().float()
+    lo_v, hi_v = gmin, gmax
+    rank_below_lo = 0  # count of elements strictly below lo_v
+    for _pass in range(8):
+        width = (hi_v - lo_v) / _NBINS
+        if width <= 0:
+            break
+        hist = torch.histc(s[(s >= lo_v) & (s <= hi_v)], bins=_NBINS, min=lo_v, max=hi_v)
+        hist = _allreduce(hist.double(), comm)
+        cum = torch.cumsum(hist, dim=0)
+        target = k - rank_below_lo
+        bin_idx = int(torch.searchsorted(cum, torch.tensor(float(target), dtype=torch.float64)).item())
+        bin_idx = min(bin_idx, _NBINS - 1)
+        in_bin = int(hist[bin_idx].item())
+        below = int(cum[bin_idx - 1].item()) if bin_idx > 0 else 0
+        new_lo = lo_v + bin_idx * width
+        new_hi = lo_v + (bin_idx + 1) * width
+        rank_below_lo += below
+        lo_v, hi_v = new_lo, new_hi
+        if not exact and in_bin <= max(1, slack):
+            return float(new_hi if bin_idx < _NBINS - 1 else hi_v)
+        if np.float32(new_lo) == np.float32(new_hi) or in_bin <= 1:
+            break
+
+    # exact finish: the k-th smallest overall is the (k - rank_below_lo)-th
+    # smallest among elements in [lo_v, hi_v]; gather those (few) and select.
+    sel = s[(s >= lo_v) & (s <= hi_v)]
+    if comm is not None:
+        vals = comm.all_gather_1d(sel.double())
+    else:
+        vals = sel.double()
+    residual = k - rank_below_lo
+    if vals.numel() == 0 or residual <= 0:
+        return float(lo_v)
+    residual = min(residual, vals.numel())
+    return float(torch.kthvalue(vals, residual).values.item())
+
+
+def verify_contamination(
+    scores: torch.Tensor,
+    threshold: float,
+    contamination: float,
+    contamination_error: float,
+    comm=None,
+) -> float:
+    """Observed-contamination check; returns the observed fraction and logs a
+    warning when outside tolerance (SharedTrainLogic.scala:211-232)."""
+    flagged = (scores.double() >= threshold).sum().to(torch.float64)
+    count = torch.tensor(float(scores.numel()), dtype=torch.float64)
+    pair = torch.stack([flagged.cpu(), count])
+    pair = _allreduce(pair, comm)
+    observed = float(pair[0].item() / max(pair[1].item(), 1.0))
+    verification_error = (
+        contamination_error if contamination_error > 0.0 else 0.01 * contamination
+    )
+    if abs(observed - contamination) > verification_error:
+        logger.warning(
+            "observed contamination %.6f deviates from expected %.6f by more "
+            "than %.6f; increase numEstimators/maxSamples or relax "
+            "contaminationError",
+            observed,
+            contamination,
+            verification_error,
+        )
+    return observed

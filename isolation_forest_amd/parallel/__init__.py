@@ -1,0 +1,3 @@
+from .comm import Comm, init_from_env
+
+__all__ = ["Comm", "init_from_env"]

@@ -1,0 +1,86 @@
+import os
+import sys
+
+import numpy as np
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+REFERENCE_RESOURCES = "/root/reference/isolation-forest/src/test/resources"
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an AMD GPU (MI355X) and the HIP extension"
+    )
+
+
+def pytest_collection_modifyitems(config, items):
+    try:
+        import torch
+
+        has_gpu = torch.cuda.is_available()
+    except Exception:
+        has_gpu = False
+    if not has_gpu:
+        skip = pytest.mark.skip(reason="no GPU available")
+        for item in items:
+            if "gpu" in item.keywords:
+                item.add_marker(skip)
+
+
+@pytest.fixture(scope="session")
+def gaussian_data():
+    """Synthetic Gaussian-mixture anomaly dataset: 4000 inliers from a tight
+    cluster mix + 120 outliers from a broad background (labels: 1=outlier).
+    Stands in for the ODDS datasets the reference tests against (no network
+    in this environment)."""
+    rs = np.random.RandomState(7)
+    d = 6
+    inl1 = rs.normal(0.0, 1.0, size=(2500, d)).astype(np.float32)
+    inl2 = (rs.normal(0.0, 0.7, size=(1500, d)) + 4.0).astype(np.float32)
+    out = rs.uniform(-12, 16, size=(120, d)).astype(np.float32)
+    X = np.concatenate([inl1, inl2, out]).astype(np.float32)
+    y = np.concatenate([np.zeros(4000), np.ones(120)])
+    perm = rs.permutation(len(X))
+    return X[perm], y[perm]
+
+
+def auroc(y_true, scores):
+    """Rank-based AUROC (no sklearn dependency)."""
+    y_true = np.asarray(y_true)
+    scores = np.asarray(scores, dtype=np.float64)
+    order = np.argsort(scores, kind="mergesort")
+    ranks = np.empty(len(scores), dtype=np.float64)
+    ranks[order] = np.arange(1, len(scores) + 1)
+    # average ranks for ties
+    sorted_scores = scores[order]
+    i = 0
+    while i < len(sorted_scores):
+        j = i
+        while j + 1 < len(sorted_scores) and sorted_scores[j + 1] == sorted_scores[i]:
+            j += 1
+        if j > i:
+            avg = (i + j) / 2.0 + 1.0
+            ranks[order[i : j + 1]] = avg
+        i = j + 1
+    pos = y_true == 1
+    n_pos = pos.sum()
+    n_neg = len(y_true) - n_pos
+    if n_pos == 0 or n_neg == 0:
+        raise ValueError("need both classes")
+    return (ranks[pos].sum() - n_pos * (n_pos + 1) / 2.0) / (n_pos * n_neg)
+
+
+@pytest.fixture(scope="session")
+def mammography():
+    """The ODDS mammography dataset from the reference checkout, when present
+    (11183 rows x 6 features + label). Skips cleanly when unavailable
+    (e.g. on a GPU box where only /root/repo is snapshotted)."""
+    path = os.path.join(REFERENCE_RESOURCES, "mammography.csv")
+    if not os.path.exists(path):
+        pytest.skip("reference mammography.csv not available")
+    raw = np.loadtxt(path, delimiter=",", dtype=np.float64)
+    X = raw[:, :-1].astype(np.float32)
+    y = raw[:, -1]
+    return X, y
