@@ -1,0 +1,221 @@
+"""Persistence tests: Avro round-trip, reference golden fixtures, format parity.
+
+Reference: IsolationForestModelWriteReadTest.scala,
+ExtendedIsolationForestModelWriteReadTest.scala.
+"""
+
+import glob
+import json
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from isolation_forest_amd import (
+    ExtendedIsolationForest,
+    ExtendedIsolationForestModel,
+    IsolationForest,
+    IsolationForestModel,
+)
+from isolation_forest_amd.persist import avro_io
+from tests.conftest import REFERENCE_RESOURCES, auroc
+
+
+@pytest.fixture(scope="module")
+def trained(gaussian_data):
+    X, _ = gaussian_data
+    model = IsolationForest(
+        numEstimators=20, contamination=0.05, contaminationError=0.0, randomSeed=11
+    ).fit(X)
+    return model, X
+
+
+@pytest.fixture(scope="module")
+def trained_ext(gaussian_data):
+    X, _ = gaussian_data
+    model = ExtendedIsolationForest(
+        numEstimators=15, contamination=0.03, randomSeed=11
+    ).fit(X)
+    return model, X
+
+
+class TestAvroContainer:
+    def test_container_roundtrip(self, tmp_path):
+        recs = [
+            {"treeID": 0, "nodeData": {"id": 0, "leftChild": 1, "rightChild": 2,
+                                       "splitAttribute": 3, "splitValue": 0.25,
+                                       "numInstances": -1}},
+            {"treeID": 0, "nodeData": {"id": 1, "leftChild": -1, "rightChild": -1,
+                                       "splitAttribute": -1, "splitValue": 0.0,
+                                       "numInstances": 7}},
+        ]
+        for codec in ("null", "deflate"):
+            p = str(tmp_path / f"t_{codec}.avro")
+            avro_io.write_container(p, avro_io.STANDARD_SCHEMA, recs, codec=codec)
+            schema, out = avro_io.read_container(p)
+            assert out == recs
+
+    def test_extended_schema_roundtrip(self, tmp_path):
+        recs = [
+            {"treeID": 1, "extendedNodeData": {
+                "id": 0, "leftChild": 1, "rightChild": 4,
+                "indices": [0, 2, 5], "weights": [0.5, -0.25, 0.125],
+                "offset": 1.0 / 3.0, "numInstances": -1}},
+            {"treeID": 1, "extendedNodeData": {
+                "id": 1, "leftChild": -1, "rightChild": -1,
+                "indices": [], "weights": [], "offset": 0.0,
+                "numInstances": 0}},
+        ]
+        p = str(tmp_path / "ext.avro")
+        avro_io.write_container(p, avro_io.EXTENDED_SCHEMA, recs, codec="deflate")
+        _, out = avro_io.read_container(p)
+        assert out == recs
+
+
+class TestModelRoundTrip:
+    def test_standard(self, trained, tmp_path):
+        model, X = trained
+        path = str(tmp_path / "model")
+        model.save(path)
+        loaded = IsolationForestModel.load(path)
+        # params, facts, threshold (IsolationForestModelWriteReadTest:41-110)
+        assert loaded.params.to_dict() == model.params.to_dict()
+        assert loaded.num_samples == model.num_samples
+        assert loaded.num_features == model.num_features
+        assert loaded.total_num_features == model.total_num_features
+        assert loaded.outlier_score_threshold == model.outlier_score_threshold
+        # exact structural equality via toString fingerprints
+        for t in range(model.forest.num_trees):
+            assert loaded.forest.tree_to_string(t) == model.forest.tree_to_string(t)
+        # identical scores and labels
+        s0 = model.transform(X)
+        s1 = loaded.transform(X)
+        assert torch.equal(s0["outlierScore"], s1["outlierScore"])
+        assert torch.equal(s0["predictedLabel"], s1["predictedLabel"])
+
+    def test_save_load_save_identical_bytes_logical(self, trained, tmp_path):
+        model, _ = trained
+        p1, p2 = str(tmp_path / "m1"), str(tmp_path / "m2")
+        model.save(p1)
+        IsolationForestModel.load(p1).save(p2)
+        _, r1 = avro_io.read_container(glob.glob(os.path.join(p1, "data", "*.avro"))[0])
+        _, r2 = avro_io.read_container(glob.glob(os.path.join(p2, "data", "*.avro"))[0])
+        assert r1 == r2
+
+    def test_extended(self, trained_ext, tmp_path):
+        model, X = trained_ext
+        path = str(tmp_path / "emodel")
+        model.save(path)
+        loaded = ExtendedIsolationForestModel.load(path)
+        assert loaded.extension_level == model.extension_level
+        for t in range(model.forest.num_trees):
+            assert loaded.forest.tree_to_string(t) == model.forest.tree_to_string(t)
+        assert torch.equal(
+            model.transform(X)["outlierScore"], loaded.transform(X)["outlierScore"]
+        )
+
+    def test_overwrite_semantics(self, trained, tmp_path):
+        model, _ = trained
+        path = str(tmp_path / "model")
+        model.save(path)
+        with pytest.raises(FileExistsError):
+            model.save(path)
+        model.write.overwrite.save(path)  # spark-style handle
+        assert os.path.exists(os.path.join(path, "metadata", "part-00000"))
+
+    def test_wrong_loader_class(self, trained, tmp_path):
+        model, _ = trained
+        path = str(tmp_path / "model")
+        model.save(path)
+        with pytest.raises(ValueError, match="wrong loader"):
+            ExtendedIsolationForestModel.load(path)
+
+    def test_metadata_shape(self, trained, tmp_path):
+        model, _ = trained
+        path = str(tmp_path / "model")
+        model.save(path)
+        meta = json.loads(open(os.path.join(path, "metadata", "part-00000")).readline())
+        assert meta["class"].endswith("IsolationForestModel")
+        for key in ("timestamp", "sparkVersion", "uid", "paramMap",
+                    "outlierScoreThreshold", "numSamples", "numFeatures",
+                    "totalNumFeatures"):
+            assert key in meta
+        assert meta["paramMap"]["numEstimators"] == 20
+
+    def test_legacy_metadata_without_total_num_features(self, trained, tmp_path):
+        # legacy models load with sentinel -1 and skip dim validation
+        # (IsolationForestModelWriteReadTest:378-460)
+        model, X = trained
+        path = str(tmp_path / "model")
+        model.save(path)
+        mp = os.path.join(path, "metadata", "part-00000")
+        meta = json.loads(open(mp).readline())
+        del meta["totalNumFeatures"]
+        open(mp, "w").write(json.dumps(meta))
+        loaded = IsolationForestModel.load(path)
+        assert loaded.total_num_features == -1
+        loaded.score(torch.from_numpy(X[:10]))  # no dim check when unknown
+
+    def test_estimator_roundtrip(self, tmp_path):
+        est = IsolationForest(numEstimators=33, contamination=0.1)
+        path = str(tmp_path / "est")
+        est.save(path)
+        loaded = IsolationForest.load(path)
+        assert loaded.getNumEstimators() == 33
+        assert loaded.getContamination() == 0.1
+
+
+needs_reference = pytest.mark.skipif(
+    not os.path.isdir(REFERENCE_RESOURCES),
+    reason="reference checkout not available",
+)
+
+
+@needs_reference
+class TestReferenceGoldenFixtures:
+    """Load the models committed to the reference repo (written by Spark,
+    snappy-coded Avro) and reproduce their golden structure exactly —
+    the backward-compat anchor (IsolationForestModelWriteReadTest:391-408)."""
+
+    def test_load_standard_golden(self):
+        path = os.path.join(REFERENCE_RESOURCES, "savedIsolationForestModel")
+        model = IsolationForestModel.load(path)
+        assert model.forest.num_trees == 100
+        assert model.num_samples == 256
+        assert model.num_features == 6
+        assert model.outlier_score_threshold == pytest.approx(0.6015323679815825)
+        # the golden file is tree 0's toString (savedIsolationForestModelTreeStructureTest)
+        golden = open(os.path.join(REFERENCE_RESOURCES, "expectedTreeStructure.txt")).read()
+        assert model.forest.tree_to_string(0) == golden
+
+    def test_load_extended_golden(self):
+        path = os.path.join(REFERENCE_RESOURCES, "savedExtendedIsolationForestModel")
+        model = ExtendedIsolationForestModel.load(path)
+        assert model.forest.num_trees == 100
+        golden = open(
+            os.path.join(REFERENCE_RESOURCES, "expectedExtendedTreeStructure.txt")
+        ).read()
+        assert model.forest.tree_to_string(0) == golden
+
+    def test_golden_model_scores_mammography(self, mammography):
+        """Our scorer over Spark's 2018 model reproduces the reference AUROC."""
+        X, y = mammography
+        path = os.path.join(REFERENCE_RESOURCES, "savedIsolationForestModel")
+        model = IsolationForestModel.load(path)
+        scores = model.score(torch.from_numpy(X)).numpy()
+        assert auroc(y, scores) == pytest.approx(0.86, abs=0.02)
+        # and the persisted contamination threshold flags ~2%
+        labels = model.labels_from_scores(torch.from_numpy(scores))
+        assert float(labels.mean()) == pytest.approx(0.02, abs=0.01)
+
+    def test_roundtrip_reference_model(self, tmp_path):
+        """save(load(reference)) preserves every node record exactly."""
+        src = os.path.join(REFERENCE_RESOURCES, "savedIsolationForestModel")
+        model = IsolationForestModel.load(src)
+        dst = str(tmp_path / "rt")
+        model.save(dst)
+        reloaded = IsolationForestModel.load(dst)
+        for t in range(model.forest.num_trees):
+            assert reloaded.forest.tree_to_string(t) == model.forest.tree_to_string(t)
+        assert reloaded.outlier_score_threshold == model.outlier_score_threshold
