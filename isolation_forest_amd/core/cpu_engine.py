@@ -45,7 +45,7 @@ from typing import Optional, Tuple
 
 import numpy as np
 
-from ..utils import rng
+from ..utils import det_math, rng
 from ..utils.math import avg_path_length
 from .forest import (
     ExtendedForest,
@@ -276,11 +276,9 @@ def build_forest(
 
 def _gaussian(seed: int, tree_id: int, counter: int) -> float:
     u1, u2 = rng.uniform2(seed, rng.P_EIF_NORMAL, np.uint32(tree_id), np.uint32(counter))
-    u1 = float(u1)
-    u2 = float(u2)
-    # Box-Muller; guard log(0)
-    r = math.sqrt(-2.0 * math.log(1.0 - u1)) if u1 < 1.0 else 0.0
-    return r * math.cos(2.0 * math.pi * u2)
+    # Box-Muller with bitwise-deterministic log/cos (utils/det_math.py):
+    # the HIP device implementation produces identical doubles.
+    return float(det_math.det_gaussian(u1, u2))
 
 
 def _draw_hyperplane(

@@ -1,0 +1,255 @@
+// Torch-extension bindings for the MI355X isolation-forest kernels.
+// Pure host glue (compiled by g++): tensor checks, LDS sizing, dispatch into
+// the hipcc-compiled launchers in forest_kernels.hip.
+
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+
+#include <vector>
+
+namespace ifa {
+
+void launch_bag_gather(bool bf16, const void* X, const int64_t* bag_idx,
+                       float* bags, int64_t T, int64_t n, int64_t d,
+                       hipStream_t stream);
+
+void launch_build_forest(const float* bags, const int32_t* feat_sub,
+                         int32_t* feat, float* value, int32_t* right,
+                         int32_t* count, int32_t* ncount,
+                         const float* leaf_lut, uint64_t seed,
+                         int32_t tree_id_offset, int32_t T, int32_t n,
+                         int32_t d, int32_t k, int32_t max_nodes,
+                         int32_t height_limit, size_t lds, hipStream_t stream);
+
+void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
+                                  int32_t* feat, float* value, int32_t* right,
+                                  int32_t* count, int32_t* ncount,
+                                  int32_t* hidx, float* hw, double* off64,
+                                  const float* leaf_lut, uint64_t seed,
+                                  int32_t tree_id_offset, int32_t T, int32_t n,
+                                  int32_t d, int32_t k, int32_t nnz,
+                                  int32_t max_nodes, int32_t height_limit,
+                                  size_t lds, hipStream_t stream);
+
+void launch_score_forest(bool bf16, bool rows_lds, const void* X,
+                         const int4* nodes, const int32_t* ncount, float* out,
+                         int64_t N, int32_t d, int32_t T, int32_t max_nodes,
+                         float fT, float c_norm, int finalize, size_t lds,
+                         int blocks, hipStream_t stream);
+
+void launch_score_extended_forest(bool bf16, bool rows_lds, bool hyper_lds,
+                                  const void* X, const int4* nodes,
+                                  const int32_t* hidx, const float* hw,
+                                  const int32_t* ncount, float* out, int64_t N,
+                                  int32_t d, int32_t T, int32_t max_nodes,
+                                  int32_t nnz, float fT, float c_norm,
+                                  int finalize, size_t lds, int blocks,
+                                  hipStream_t stream);
+
+}  // namespace ifa
+
+namespace {
+
+constexpr size_t kMaxLds = 160 * 1024;
+
+#define CHECK_CUDA(x) TORCH_CHECK((x).is_cuda(), #x " must be a GPU tensor")
+#define CHECK_CONTIG(x) TORCH_CHECK((x).is_contiguous(), #x " must be contiguous")
+
+hipStream_t current_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+inline size_t align16(size_t x) { return (x + 15) & ~(size_t)15; }
+
+bool is_bf16(const torch::Tensor& t) {
+  return t.scalar_type() == torch::kBFloat16;
+}
+
+void check_x(const torch::Tensor& X) {
+  TORCH_CHECK(X.dim() == 2, "X must be [N, d]");
+  TORCH_CHECK(
+      X.scalar_type() == torch::kFloat32 || X.scalar_type() == torch::kBFloat16,
+      "X must be float32 or bfloat16");
+}
+
+}  // namespace
+
+torch::Tensor bag_gather(torch::Tensor X, torch::Tensor bag_idx) {
+  CHECK_CUDA(X);
+  CHECK_CONTIG(X);
+  CHECK_CUDA(bag_idx);
+  CHECK_CONTIG(bag_idx);
+  check_x(X);
+  TORCH_CHECK(bag_idx.dim() == 2 && bag_idx.scalar_type() == torch::kInt64,
+              "bag_idx must be int64 [T, n]");
+  int64_t T = bag_idx.size(0), n = bag_idx.size(1), d = X.size(1);
+  auto bags = torch::empty({T, n, d}, X.options().dtype(torch::kFloat32));
+  ifa::launch_bag_gather(is_bf16(X), X.data_ptr(),
+                         bag_idx.data_ptr<int64_t>(), bags.data_ptr<float>(),
+                         T, n, d, current_stream());
+  return bags;
+}
+
+std::vector<torch::Tensor> build_forest(torch::Tensor bags,
+                                        torch::Tensor feat_sub, int64_t seed,
+                                        int64_t tree_id_offset,
+                                        torch::Tensor leaf_lut,
+                                        int64_t max_nodes,
+                                        int64_t height_limit) {
+  CHECK_CUDA(bags);
+  CHECK_CONTIG(bags);
+  CHECK_CUDA(feat_sub);
+  CHECK_CONTIG(feat_sub);
+  CHECK_CUDA(leaf_lut);
+  CHECK_CONTIG(leaf_lut);
+  TORCH_CHECK(bags.dim() == 3 && bags.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(feat_sub.scalar_type() == torch::kInt32);
+  int64_t T = bags.size(0), n = bags.size(1), d = bags.size(2);
+  int64_t k = feat_sub.size(1);
+  TORCH_CHECK(n <= 16384, "GPU build supports maxSamples <= 16384");
+  TORCH_CHECK(max_nodes <= 32767, "max_nodes must fit int16 patch ids");
+
+  auto opts = bags.options().dtype(torch::kInt32);
+  auto feat = torch::empty({T, max_nodes}, opts);
+  auto value = torch::empty({T, max_nodes}, bags.options());
+  auto right = torch::full({T, max_nodes}, -1, opts);
+  auto count = torch::empty({T, max_nodes}, opts);
+  auto ncount = torch::empty({T}, opts);
+
+  size_t lds = align16(4 * n + 2 * k) + (height_limit + 8) * 8;
+  TORCH_CHECK(lds <= kMaxLds, "build LDS request too large");
+  ifa::launch_build_forest(
+      bags.data_ptr<float>(), feat_sub.data_ptr<int32_t>(),
+      feat.data_ptr<int32_t>(), value.data_ptr<float>(),
+      right.data_ptr<int32_t>(), count.data_ptr<int32_t>(),
+      ncount.data_ptr<int32_t>(), leaf_lut.data_ptr<float>(), (uint64_t)seed,
+      (int32_t)tree_id_offset, (int32_t)T, (int32_t)n, (int32_t)d, (int32_t)k,
+      (int32_t)max_nodes, (int32_t)height_limit, lds, current_stream());
+  return {feat, value, right, count, ncount};
+}
+
+std::vector<torch::Tensor> build_extended_forest(
+    torch::Tensor bags, torch::Tensor feat_sub, int64_t seed,
+    int64_t tree_id_offset, torch::Tensor leaf_lut, int64_t nnz,
+    int64_t max_nodes, int64_t height_limit) {
+  CHECK_CUDA(bags);
+  CHECK_CONTIG(bags);
+  CHECK_CUDA(feat_sub);
+  CHECK_CONTIG(feat_sub);
+  CHECK_CUDA(leaf_lut);
+  int64_t T = bags.size(0), n = bags.size(1), d = bags.size(2);
+  int64_t k = feat_sub.size(1);
+  TORCH_CHECK(n <= 16384, "GPU build supports maxSamples <= 16384");
+  TORCH_CHECK(max_nodes <= 32767);
+  TORCH_CHECK(nnz >= 1 && nnz <= k);
+
+  auto opts = bags.options().dtype(torch::kInt32);
+  auto feat = torch::empty({T, max_nodes}, opts);
+  auto value = torch::empty({T, max_nodes}, bags.options());
+  auto right = torch::full({T, max_nodes}, -1, opts);
+  auto count = torch::empty({T, max_nodes}, opts);
+  auto ncount = torch::empty({T}, opts);
+  auto hidx = torch::zeros({T, max_nodes, nnz}, opts);
+  auto hw = torch::zeros({T, max_nodes, nnz}, bags.options());
+  auto off64 =
+      torch::zeros({T, max_nodes}, bags.options().dtype(torch::kFloat64));
+
+  size_t lds =
+      align16(4 * n + 2 * k) + align16(8 * nnz) + (height_limit + 8) * 8 + 32;
+  TORCH_CHECK(lds <= kMaxLds, "build LDS request too large");
+  ifa::launch_build_extended_forest(
+      bags.data_ptr<float>(), feat_sub.data_ptr<int32_t>(),
+      feat.data_ptr<int32_t>(), value.data_ptr<float>(),
+      right.data_ptr<int32_t>(), count.data_ptr<int32_t>(),
+      ncount.data_ptr<int32_t>(), hidx.data_ptr<int32_t>(),
+      hw.data_ptr<float>(), off64.data_ptr<double>(),
+      leaf_lut.data_ptr<float>(), (uint64_t)seed, (int32_t)tree_id_offset,
+      (int32_t)T, (int32_t)n, (int32_t)d, (int32_t)k, (int32_t)nnz,
+      (int32_t)max_nodes, (int32_t)height_limit, lds, current_stream());
+  return {feat, value, right, count, ncount, hidx, hw, off64};
+}
+
+torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_aos,
+                           torch::Tensor ncount, double c_norm,
+                           bool finalize) {
+  CHECK_CUDA(X);
+  CHECK_CONTIG(X);
+  CHECK_CUDA(nodes_aos);
+  CHECK_CONTIG(nodes_aos);
+  CHECK_CUDA(ncount);
+  check_x(X);
+  TORCH_CHECK(nodes_aos.dim() == 3 && nodes_aos.size(2) == 4 &&
+                  nodes_aos.scalar_type() == torch::kInt32,
+              "nodes_aos must be int32 [T, max_nodes, 4]");
+  int64_t N = X.size(0), d = X.size(1);
+  int64_t T = nodes_aos.size(0), max_nodes = nodes_aos.size(1);
+  auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
+  if (N == 0) return out;
+
+  size_t node_bytes = (size_t)max_nodes * 16;
+  size_t row_bytes = (size_t)256 * (d + 1) * 4;
+  bool rows_lds = (d % 4 == 0) && (node_bytes + row_bytes <= 128 * 1024);
+  size_t lds = node_bytes + (rows_lds ? row_bytes : 0);
+  TORCH_CHECK(node_bytes <= kMaxLds, "tree too large for LDS staging");
+
+  int blocks = (int)std::min<int64_t>((N + 255) / 256, 4096);
+  ifa::launch_score_forest(is_bf16(X), rows_lds, X.data_ptr(),
+                           (const int4*)nodes_aos.data_ptr<int32_t>(),
+                           ncount.data_ptr<int32_t>(), out.data_ptr<float>(),
+                           N, (int32_t)d, (int32_t)T, (int32_t)max_nodes,
+                           (float)T, (float)c_norm, finalize ? 1 : 0, lds,
+                           blocks, current_stream());
+  return out;
+}
+
+torch::Tensor score_extended_forest(torch::Tensor X, torch::Tensor nodes_aos,
+                                    torch::Tensor hidx, torch::Tensor hw,
+                                    torch::Tensor ncount, double c_norm,
+                                    bool finalize) {
+  CHECK_CUDA(X);
+  CHECK_CONTIG(X);
+  CHECK_CUDA(nodes_aos);
+  CHECK_CONTIG(nodes_aos);
+  CHECK_CUDA(hidx);
+  CHECK_CONTIG(hidx);
+  CHECK_CUDA(hw);
+  CHECK_CONTIG(hw);
+  check_x(X);
+  int64_t N = X.size(0), d = X.size(1);
+  int64_t T = nodes_aos.size(0), max_nodes = nodes_aos.size(1);
+  int64_t nnz = hidx.size(2);
+  auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
+  if (N == 0) return out;
+
+  size_t node_bytes = (size_t)max_nodes * 16;
+  size_t hyper_bytes = (size_t)max_nodes * nnz * 8;
+  size_t row_bytes = (size_t)256 * (d + 1) * 4;
+  bool hyper_lds = node_bytes + hyper_bytes <= 96 * 1024;
+  bool rows_lds =
+      (d % 4 == 0) &&
+      (node_bytes + (hyper_lds ? hyper_bytes : 0) + row_bytes <= 144 * 1024);
+  size_t lds =
+      node_bytes + (hyper_lds ? hyper_bytes : 0) + (rows_lds ? row_bytes : 0);
+  TORCH_CHECK(node_bytes <= kMaxLds, "tree too large for LDS staging");
+
+  int blocks = (int)std::min<int64_t>((N + 255) / 256, 4096);
+  ifa::launch_score_extended_forest(
+      is_bf16(X), rows_lds, hyper_lds, X.data_ptr(),
+      (const int4*)nodes_aos.data_ptr<int32_t>(), hidx.data_ptr<int32_t>(),
+      hw.data_ptr<float>(), ncount.data_ptr<int32_t>(), out.data_ptr<float>(),
+      N, (int32_t)d, (int32_t)T, (int32_t)max_nodes, (int32_t)nnz, (float)T,
+      (float)c_norm, finalize ? 1 : 0, lds, blocks, current_stream());
+  return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bag_gather", &bag_gather, "gather per-tree bags (K9/K10)");
+  m.def("build_forest", &build_forest, "build standard iTrees (K1/K2/K11)");
+  m.def("build_extended_forest", &build_extended_forest,
+        "build extended iTrees (K3-K5)");
+  m.def("score_forest", &score_forest, "batched path-length scoring (K6)");
+  m.def("score_extended_forest", &score_extended_forest,
+        "batched EIF scoring (K7)");
+  m.attr("WAVE") = 64;
+}

@@ -1,0 +1,716 @@
+// MI355X (gfx950, CDNA4) Isolation Forest kernels.
+//
+// These are the hand-written HIP implementations of the reference's JVM hot
+// loops:
+//  * build_forest_kernel          — K1 (min/max scan + constant-feature
+//    retry), K2 (stable row partition), K11 (per-node feature Fisher-Yates):
+//    one 64-lane wavefront per tree, row-index permutation in LDS, ballot/
+//    popcount stable partition, shfl-xor min/max reductions.
+//  * build_extended_forest_kernel — K3-K5 (EIF Gaussian hyperplane draw,
+//    L2 normalize, per-coordinate min/max + offset, dot-product partition).
+//  * score_forest_kernel          — K6/K8 (batched path-length traversal):
+//    one thread per row, the current tree's nodes staged in LDS as 16-byte
+//    AoS records (one ds_read_b128 per visit), rows optionally staged in
+//    LDS (padded to kill bank conflicts), float32 path-sum accumulation in
+//    tree order (bitwise == the CPU oracle), leaf c(n) terms precomputed.
+//  * score_extended_forest_kernel — K7 (EIF traversal, sparse dot per level).
+//  * bag_gather_kernel            — K9/K10 device side: gather sampled rows
+//    into per-tree bags (indices drawn host-side by the same Philox).
+//
+// PRECISION CONTRACT: see core/cpu_engine.py. All randomness: philox.h;
+// EIF gaussians: det_math.h. Wavefront size is 64 (CDNA4) and is hard-coded.
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#include "det_math.h"
+#include "philox.h"
+
+#define WAVE 64
+#define LANE_MASK_LT(lane) ((1ull << (lane)) - 1ull)
+
+namespace ifa {
+
+// ---------------------------------------------------------------------------
+// helpers
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ float wave_reduce_min(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fminf(v, __shfl_xor(v, off, WAVE));
+  return v;
+}
+
+__device__ __forceinline__ float wave_reduce_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, WAVE));
+  return v;
+}
+
+// split point: float64 draw rounded to float32, bumped off the left edge
+// (cpu_engine._split_value32)
+__device__ __forceinline__ float split_value32(float fmin, float fmax, double u) {
+  double s64 = (double)fmin + u * ((double)fmax - (double)fmin);
+  float s32 = (float)s64;
+  if (!(fmin < s32)) s32 = nextafterf(fmin, __builtin_inff());
+  return s32;
+}
+
+struct SegEntry {
+  uint16_t start, end;
+  int16_t height;
+  int16_t patch;  // node whose `right` points at the node this entry emits
+};
+
+// ---------------------------------------------------------------------------
+// bag gather: bags[T][n][d] (f32) <- X[N][d] (f32 or bf16) at idx[T][n]
+// ---------------------------------------------------------------------------
+
+template <typename XT>
+__device__ __forceinline__ float load_feat(const XT* X, int64_t off);
+
+template <>
+__device__ __forceinline__ float load_feat<float>(const float* X, int64_t off) {
+  return X[off];
+}
+template <>
+__device__ __forceinline__ float load_feat<uint16_t>(const uint16_t* X, int64_t off) {
+  union { uint32_t u; float f; } cv;
+  cv.u = ((uint32_t)X[off]) << 16;  // bf16 -> f32 (exact)
+  return cv.f;
+}
+
+template <typename XT>
+__global__ void bag_gather_kernel(const XT* __restrict__ X,
+                                  const int64_t* __restrict__ bag_idx,
+                                  float* __restrict__ bags, int64_t T, int64_t n,
+                                  int64_t d) {
+  int64_t total = T * n * d;
+  for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; g < total;
+       g += (int64_t)gridDim.x * blockDim.x) {
+    int64_t j = g % d;
+    int64_t ts = g / d;  // t*n + s
+    int64_t row = bag_idx[ts];
+    bags[g] = load_feat<XT>(X, row * d + j);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// standard build: one wavefront per tree
+// ---------------------------------------------------------------------------
+
+extern __shared__ __attribute__((aligned(16))) char smem[];
+
+__global__ void __launch_bounds__(WAVE) build_forest_kernel(
+    const float* __restrict__ bags,        // [T][n][d]
+    const int32_t* __restrict__ feat_sub,  // [T][k] sorted global feature ids
+    int32_t* __restrict__ out_feat,        // [T][max_nodes]
+    float* __restrict__ out_value,         // [T][max_nodes]
+    int32_t* __restrict__ out_right,       // [T][max_nodes]
+    int32_t* __restrict__ out_count,       // [T][max_nodes] (leaf counts; -1)
+    int32_t* __restrict__ out_ncount,      // [T]
+    const float* __restrict__ leaf_lut,    // [n+1]: c(m) float32
+    uint64_t seed, int32_t tree_id_offset, int32_t n, int32_t d, int32_t k,
+    int32_t max_nodes, int32_t height_limit) {
+  const int t = blockIdx.x;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const uint32_t gtree = (uint32_t)(t + tree_id_offset);
+
+  uint16_t* idx = (uint16_t*)smem;               // [n]
+  uint16_t* tmp = idx + n;                       // [n]
+  uint16_t* avail = tmp + n;                     // [k]
+  SegEntry* stack = (SegEntry*)(((uintptr_t)(avail + k) + 15) & ~15ull);
+
+  const float* bag = bags + (int64_t)t * n * d;
+  int32_t* feat = out_feat + (int64_t)t * max_nodes;
+  float* value = out_value + (int64_t)t * max_nodes;
+  int32_t* right = out_right + (int64_t)t * max_nodes;
+  int32_t* count = out_count + (int64_t)t * max_nodes;
+  const int32_t* features = feat_sub + (int64_t)t * k;
+
+  for (int i = lane; i < n; i += WAVE) idx[i] = (uint16_t)i;
+  if (lane == 0) stack[0] = SegEntry{0, (uint16_t)n, 0, -1};
+  __syncthreads();
+
+  int sp = 1;
+  int next_id = 0;
+  while (sp > 0) {
+    SegEntry e = stack[--sp];
+    const int node = next_id++;
+    if (e.patch >= 0 && lane == 0) right[e.patch] = node;
+    const int start = e.start, end = e.end, m = end - start;
+
+    if (m <= 1 || e.height >= height_limit) {
+      if (lane == 0) {
+        feat[node] = -1;
+        value[node] = leaf_lut[m];
+        count[node] = m;
+      }
+      continue;
+    }
+
+    // ---- feature selection with constant-feature retry (K1 + K11) ----
+    for (int j = lane; j < k; j += WAVE) avail[j] = (uint16_t)features[j];
+    __syncthreads();
+    int found = -1;
+    float fmin = 0.f, fmax = 0.f;
+    for (int j = 0; j < k; ++j) {
+      uint32_t r = rng_below(seed, P_FEATSEL, gtree, (uint32_t)node,
+                             (uint32_t)(k - j), (uint32_t)j);
+      int tsel = j + (int)r;
+      if (lane == 0) {
+        uint16_t a = avail[j];
+        avail[j] = avail[tsel];
+        avail[tsel] = a;
+      }
+      __syncthreads();
+      const int f = avail[j];
+      float lo = __builtin_inff(), hi = -__builtin_inff();
+      for (int i = start + lane; i < end; i += WAVE) {
+        float v = bag[(int64_t)idx[i] * d + f];
+        lo = fminf(lo, v);
+        hi = fmaxf(hi, v);
+      }
+      lo = wave_reduce_min(lo);
+      hi = wave_reduce_max(hi);
+      if (lo < hi) {
+        found = f;
+        fmin = lo;
+        fmax = hi;
+        break;
+      }
+    }
+    if (found < 0) {
+      if (lane == 0) {
+        feat[node] = -1;
+        value[node] = leaf_lut[m];
+        count[node] = m;
+      }
+      continue;
+    }
+
+    const double u = rng_uniform(seed, P_SPLIT, gtree, (uint32_t)node);
+    const float s32 = split_value32(fmin, fmax, u);
+
+    // ---- stable partition via ballot/popcount (K2) ----
+    int cntL = 0;
+    for (int base = start; base < end; base += WAVE) {
+      const int i = base + lane;
+      const bool valid = i < end;
+      float v = valid ? bag[(int64_t)idx[i] * d + found] : 0.f;
+      const bool p = valid && (v < s32);
+      cntL += __popcll(__ballot(p));
+    }
+    int runL = 0, runR = 0;
+    for (int base = start; base < end; base += WAVE) {
+      const int i = base + lane;
+      const bool valid = i < end;
+      uint16_t my = valid ? idx[i] : (uint16_t)0;
+      float v = valid ? bag[(int64_t)my * d + found] : 0.f;
+      const bool p = valid && (v < s32);
+      const uint64_t maskL = __ballot(p);
+      const uint64_t maskR = __ballot(valid && !p);
+      if (p)
+        tmp[start + runL + __popcll(maskL & LANE_MASK_LT(lane))] = my;
+      else if (valid)
+        tmp[start + cntL + runR + __popcll(maskR & LANE_MASK_LT(lane))] = my;
+      runL += __popcll(maskL);
+      runR += __popcll(maskR);
+    }
+    __syncthreads();
+    for (int i = start + lane; i < end; i += WAVE) idx[i] = tmp[i];
+    __syncthreads();
+
+    if (lane == 0) {
+      feat[node] = found;
+      value[node] = s32;
+      count[node] = -1;
+      // pre-order: push right, then left (left pops first = node+1)
+      stack[sp] = SegEntry{(uint16_t)(start + cntL), (uint16_t)end,
+                           (int16_t)(e.height + 1), (int16_t)node};
+      stack[sp + 1] = SegEntry{(uint16_t)start, (uint16_t)(start + cntL),
+                               (int16_t)(e.height + 1), (int16_t)-1};
+    }
+    __syncthreads();
+    sp += 2;
+  }
+  if (lane == 0) out_ncount[t] = next_id;
+}
+
+// ---------------------------------------------------------------------------
+// extended build: one wavefront per tree
+// ---------------------------------------------------------------------------
+
+__global__ void __launch_bounds__(WAVE) build_extended_forest_kernel(
+    const float* __restrict__ bags, const int32_t* __restrict__ feat_sub,
+    int32_t* __restrict__ out_feat, float* __restrict__ out_value,
+    int32_t* __restrict__ out_right, int32_t* __restrict__ out_count,
+    int32_t* __restrict__ out_ncount,
+    int32_t* __restrict__ out_hidx,    // [T][max_nodes][nnz]
+    float* __restrict__ out_hw,        // [T][max_nodes][nnz]
+    double* __restrict__ out_off64,    // [T][max_nodes]
+    const float* __restrict__ leaf_lut, uint64_t seed, int32_t tree_id_offset,
+    int32_t n, int32_t d, int32_t k, int32_t nnz, int32_t max_nodes,
+    int32_t height_limit) {
+  const int t = blockIdx.x;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const uint32_t gtree = (uint32_t)(t + tree_id_offset);
+
+  uint16_t* idx = (uint16_t*)smem;   // [n]
+  uint16_t* tmp = idx + n;           // [n]
+  uint16_t* avail = tmp + n;         // [k]
+  int32_t* coords =                  // [nnz]
+      (int32_t*)(((uintptr_t)(avail + k) + 15) & ~15ull);
+  float* wf = (float*)(coords + nnz);  // [nnz]
+  SegEntry* stack = (SegEntry*)(((uintptr_t)(wf + nnz) + 15) & ~15ull);
+
+  const float* bag = bags + (int64_t)t * n * d;
+  int32_t* feat = out_feat + (int64_t)t * max_nodes;
+  float* value = out_value + (int64_t)t * max_nodes;
+  int32_t* right = out_right + (int64_t)t * max_nodes;
+  int32_t* count = out_count + (int64_t)t * max_nodes;
+  int32_t* hidx = out_hidx + (int64_t)t * max_nodes * nnz;
+  float* hw = out_hw + (int64_t)t * max_nodes * nnz;
+  double* off64p = out_off64 + (int64_t)t * max_nodes;
+  const int32_t* features = feat_sub + (int64_t)t * k;
+
+  for (int i = lane; i < n; i += WAVE) idx[i] = (uint16_t)i;
+  if (lane == 0) stack[0] = SegEntry{0, (uint16_t)n, 0, -1};
+  __syncthreads();
+
+  int sp = 1;
+  int next_id = 0;
+  while (sp > 0) {
+    SegEntry e = stack[--sp];
+    const int node = next_id++;
+    if (e.patch >= 0 && lane == 0) right[e.patch] = node;
+    const int start = e.start, end = e.end, m = end - start;
+
+    if (m <= 1 || e.height >= height_limit) {
+      if (lane == 0) {
+        feat[node] = -1;
+        value[node] = leaf_lut[m];
+        count[node] = m;
+      }
+      continue;
+    }
+
+    // ---- coordinate subset: Fisher-Yates over the tree's subspace ----
+    for (int j = lane; j < k; j += WAVE) avail[j] = (uint16_t)features[j];
+    __syncthreads();
+    for (int j = 0; j < nnz; ++j) {
+      uint32_t r = rng_below(seed, P_EIF_COORD, gtree, (uint32_t)node,
+                             (uint32_t)(k - j), (uint32_t)j);
+      if (lane == 0) {
+        int tsel = j + (int)r;
+        uint16_t a = avail[j];
+        avail[j] = avail[tsel];
+        avail[tsel] = a;
+      }
+      __syncthreads();
+    }
+    // sort the first nnz ascending (lane 0, insertion sort; nnz <= ~128)
+    if (lane == 0) {
+      for (int a = 1; a < nnz; ++a) {
+        uint16_t key = avail[a];
+        int b = a - 1;
+        while (b >= 0 && avail[b] > key) {
+          avail[b + 1] = avail[b];
+          --b;
+        }
+        avail[b + 1] = key;
+      }
+      for (int j = 0; j < nnz; ++j) coords[j] = (int32_t)avail[j];
+    }
+    __syncthreads();
+
+    // ---- Gaussian weights per sorted slot (det Box-Muller), parallel ----
+    const uint32_t base = (uint32_t)node * 4096u;
+    for (int j = lane; j < nnz; j += WAVE) {
+      double u1, u2;
+      rng_uniform2(seed, P_EIF_NORMAL, gtree, base + (uint32_t)j, &u1, &u2);
+      wf[j] = (float)det_gaussian(u1, u2);
+    }
+    __syncthreads();
+    // sequential float32 norm (order == CPU oracle)
+    float acc = 0.f;
+    for (int j = 0; j < nnz; ++j) acc = __fadd_rn(acc, __fmul_rn(wf[j], wf[j]));
+    if (acc <= 0.f) {  // theoretical zero-norm leaf
+      if (lane == 0) {
+        feat[node] = -1;
+        value[node] = leaf_lut[m];
+        count[node] = m;
+      }
+      continue;
+    }
+    const float norm = __fsqrt_rn(acc);
+    for (int j = lane; j < nnz; j += WAVE) wf[j] = __fdiv_rn(wf[j], norm);
+    __syncthreads();
+
+    // ---- intercepts per sorted coordinate + offset (float64) ----
+    double off64 = 0.0;
+    for (int j = 0; j < nnz; ++j) {
+      const int c = coords[j];
+      float lo = __builtin_inff(), hi = -__builtin_inff();
+      for (int i = start + lane; i < end; i += WAVE) {
+        float v = bag[(int64_t)idx[i] * d + c];
+        lo = fminf(lo, v);
+        hi = fmaxf(hi, v);
+      }
+      lo = wave_reduce_min(lo);
+      hi = wave_reduce_max(hi);
+      const double u =
+          rng_uniform(seed, P_EIF_INTERCEPT, gtree, base + (uint32_t)j);
+      const double intercept = (double)lo + u * ((double)hi - (double)lo);
+      off64 += (double)wf[j] * intercept;
+    }
+    const float off32 = (float)off64;
+
+    // ---- partition by hyperplane dot (no retry; empty sides allowed) ----
+    int cntL = 0;
+    for (int basei = start; basei < end; basei += WAVE) {
+      const int i = basei + lane;
+      const bool valid = i < end;
+      float dot = 0.f;
+      if (valid) {
+        const float* rowp = bag + (int64_t)idx[i] * d;
+        for (int j = 0; j < nnz; ++j)
+          dot = __fadd_rn(dot, __fmul_rn(wf[j], rowp[coords[j]]));
+      }
+      cntL += __popcll(__ballot(valid && (dot < off32)));
+    }
+    int runL = 0, runR = 0;
+    for (int basei = start; basei < end; basei += WAVE) {
+      const int i = basei + lane;
+      const bool valid = i < end;
+      uint16_t my = valid ? idx[i] : (uint16_t)0;
+      float dot = 0.f;
+      if (valid) {
+        const float* rowp = bag + (int64_t)my * d;
+        for (int j = 0; j < nnz; ++j)
+          dot = __fadd_rn(dot, __fmul_rn(wf[j], rowp[coords[j]]));
+      }
+      const bool p = valid && (dot < off32);
+      const uint64_t maskL = __ballot(p);
+      const uint64_t maskR = __ballot(valid && !p);
+      if (p)
+        tmp[start + runL + __popcll(maskL & LANE_MASK_LT(lane))] = my;
+      else if (valid)
+        tmp[start + cntL + runR + __popcll(maskR & LANE_MASK_LT(lane))] = my;
+      runL += __popcll(maskL);
+      runR += __popcll(maskR);
+    }
+    __syncthreads();
+    for (int i = start + lane; i < end; i += WAVE) idx[i] = tmp[i];
+    __syncthreads();
+
+    if (lane == 0) {
+      feat[node] = nnz;
+      value[node] = off32;
+      off64p[node] = off64;
+      count[node] = -1;
+      for (int j = 0; j < nnz; ++j) {
+        hidx[(int64_t)node * nnz + j] = coords[j];
+        hw[(int64_t)node * nnz + j] = wf[j];
+      }
+      stack[sp] = SegEntry{(uint16_t)(start + cntL), (uint16_t)end,
+                           (int16_t)(e.height + 1), (int16_t)node};
+      stack[sp + 1] = SegEntry{(uint16_t)start, (uint16_t)(start + cntL),
+                               (int16_t)(e.height + 1), (int16_t)-1};
+    }
+    __syncthreads();
+    sp += 2;
+  }
+  if (lane == 0) out_ncount[t] = next_id;
+}
+
+// ---------------------------------------------------------------------------
+// standard scoring: the throughput kernel (K6)
+// ---------------------------------------------------------------------------
+
+// node AoS record (16 B): {feat|-1, value_bits(f32), right, unused}
+// one ds_read_b128 per visit when staged in LDS.
+
+template <typename XT, bool ROWS_LDS>
+__global__ void __launch_bounds__(256) score_forest_kernel(
+    const XT* __restrict__ X,          // [N][d]
+    const int4* __restrict__ nodes,    // [T][max_nodes] AoS
+    const int32_t* __restrict__ ncnt,  // [T]
+    float* __restrict__ out,           // [N] (path sum or score)
+    int64_t N, int32_t d, int32_t T, int32_t max_nodes, float fT,
+    float c_norm, int32_t finalize) {
+  const int tid = threadIdx.x;
+  const int dpad = d + 1;
+
+  int4* tree_lds = (int4*)smem;  // [max_nodes]
+  float* rows = (float*)(tree_lds + max_nodes);  // [256][dpad] if ROWS_LDS
+
+  for (int64_t block_row0 = (int64_t)blockIdx.x * 256; block_row0 < N;
+       block_row0 += (int64_t)gridDim.x * 256) {
+    const int64_t my_row = block_row0 + tid;
+    const int rows_here = (int)min((int64_t)256, N - block_row0);
+
+    if (ROWS_LDS) {
+      __syncthreads();  // previous iteration's reads done
+      const int64_t total = (int64_t)rows_here * d;
+      for (int64_t g = tid; g < total; g += 256) {
+        const int r = (int)(g / d), c = (int)(g % d);
+        rows[r * dpad + c] = load_feat<XT>(X, (block_row0 + r) * d + c);
+      }
+      __syncthreads();
+    }
+
+    float path_sum = 0.f;
+    for (int t = 0; t < T; ++t) {
+      // stage tree t into LDS (coop)
+      __syncthreads();
+      const int nc = ncnt[t];
+      const int4* src = nodes + (int64_t)t * max_nodes;
+      for (int i = tid; i < nc; i += 256) tree_lds[i] = src[i];
+      __syncthreads();
+
+      if (my_row < N) {
+        int node = 0;
+        int depth = 0;
+        float leaf = 0.f;
+        while (true) {
+          const int4 nd = tree_lds[node];
+          if (nd.x < 0) {
+            leaf = __int_as_float(nd.y);
+            break;
+          }
+          float xv;
+          if (ROWS_LDS)
+            xv = rows[tid * dpad + nd.x];
+          else
+            xv = load_feat<XT>(X, my_row * d + nd.x);
+          node = (xv < __int_as_float(nd.y)) ? node + 1 : nd.z;
+          ++depth;
+        }
+        // oracle order (cpu_engine.path_lengths): per tree
+        // total = f32(total + f32(depth + leaf))
+        path_sum = __fadd_rn(path_sum, __fadd_rn((float)depth, leaf));
+      }
+    }
+    if (my_row < N) {
+      if (finalize) {
+        const float mean32 = __fdiv_rn(path_sum, fT);
+        const double ratio = (double)mean32 / (double)c_norm;
+        out[my_row] = (float)exp2(-ratio);
+      } else {
+        out[my_row] = path_sum;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// extended scoring (K7)
+// ---------------------------------------------------------------------------
+
+template <typename XT, bool ROWS_LDS, bool HYPER_LDS>
+__global__ void __launch_bounds__(256) score_extended_forest_kernel(
+    const XT* __restrict__ X, const int4* __restrict__ nodes,
+    const int32_t* __restrict__ hidx_g,  // [T][max_nodes][nnz]
+    const float* __restrict__ hw_g,      // [T][max_nodes][nnz]
+    const int32_t* __restrict__ ncnt, float* __restrict__ out, int64_t N,
+    int32_t d, int32_t T, int32_t max_nodes, int32_t nnz, float fT,
+    float c_norm, int32_t finalize) {
+  const int tid = threadIdx.x;
+  const int dpad = d + 1;
+
+  int4* tree_lds = (int4*)smem;                       // [max_nodes]
+  int32_t* hidx_lds = (int32_t*)(tree_lds + max_nodes);  // [max_nodes*nnz]
+  float* hw_lds = (float*)(hidx_lds + (HYPER_LDS ? max_nodes * nnz : 0));
+  float* rows = hw_lds + (HYPER_LDS ? max_nodes * nnz : 0);
+
+  for (int64_t block_row0 = (int64_t)blockIdx.x * 256; block_row0 < N;
+       block_row0 += (int64_t)gridDim.x * 256) {
+    const int64_t my_row = block_row0 + tid;
+    const int rows_here = (int)min((int64_t)256, N - block_row0);
+
+    if (ROWS_LDS) {
+      __syncthreads();
+      const int64_t total = (int64_t)rows_here * d;
+      for (int64_t g = tid; g < total; g += 256) {
+        const int r = (int)(g / d), c = (int)(g % d);
+        rows[r * dpad + c] = load_feat<XT>(X, (block_row0 + r) * d + c);
+      }
+      __syncthreads();
+    }
+
+    float path_sum = 0.f;
+    for (int t = 0; t < T; ++t) {
+      __syncthreads();
+      const int nc = ncnt[t];
+      const int4* src = nodes + (int64_t)t * max_nodes;
+      for (int i = tid; i < nc; i += 256) tree_lds[i] = src[i];
+      if (HYPER_LDS) {
+        const int64_t hbase = (int64_t)t * max_nodes * nnz;
+        for (int i = tid; i < nc * nnz; i += 256) {
+          hidx_lds[i] = hidx_g[hbase + i];
+          hw_lds[i] = hw_g[hbase + i];
+        }
+      }
+      __syncthreads();
+
+      if (my_row < N) {
+        const int64_t hbase = (int64_t)t * max_nodes * nnz;
+        int node = 0;
+        int depth = 0;
+        float leaf = 0.f;
+        while (true) {
+          const int4 nd = tree_lds[node];
+          if (nd.x < 0) {
+            leaf = __int_as_float(nd.y);
+            break;
+          }
+          float dot = 0.f;
+          if (HYPER_LDS) {
+            const int32_t* ci = hidx_lds + node * nnz;
+            const float* cw = hw_lds + node * nnz;
+            for (int j = 0; j < nnz; ++j) {
+              float xv = ROWS_LDS ? rows[tid * dpad + ci[j]]
+                                  : load_feat<XT>(X, my_row * d + ci[j]);
+              dot = __fadd_rn(dot, __fmul_rn(cw[j], xv));
+            }
+          } else {
+            const int32_t* ci = hidx_g + hbase + (int64_t)node * nnz;
+            const float* cw = hw_g + hbase + (int64_t)node * nnz;
+            for (int j = 0; j < nnz; ++j) {
+              float xv = ROWS_LDS ? rows[tid * dpad + ci[j]]
+                                  : load_feat<XT>(X, my_row * d + ci[j]);
+              dot = __fadd_rn(dot, __fmul_rn(cw[j], xv));
+            }
+          }
+          node = (dot < __int_as_float(nd.y)) ? node + 1 : nd.z;
+          ++depth;
+        }
+        path_sum = __fadd_rn(path_sum, __fadd_rn((float)depth, leaf));
+      }
+    }
+    if (my_row < N) {
+      if (finalize) {
+        const float mean32 = __fdiv_rn(path_sum, fT);
+        const double ratio = (double)mean32 / (double)c_norm;
+        out[my_row] = (float)exp2(-ratio);
+      } else {
+        out[my_row] = path_sum;
+      }
+    }
+  }
+}
+
+}  // namespace ifa
+
+// ---------------------------------------------------------------------------
+// host-side launchers (bindings.cpp is compiled by g++, so every kernel
+// launch lives in this hipcc TU)
+// ---------------------------------------------------------------------------
+
+namespace ifa {
+
+static inline void raise_lds(const void* func, size_t bytes) {
+  if (bytes > 64 * 1024)
+    (void)hipFuncSetAttribute(func, hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)bytes);
+}
+
+void launch_bag_gather(bool bf16, const void* X, const int64_t* bag_idx,
+                       float* bags, int64_t T, int64_t n, int64_t d,
+                       hipStream_t stream) {
+  int64_t total = T * n * d;
+  int threads = 256;
+  int blocks = (int)((total + threads - 1) / threads);
+  if (blocks > 4096) blocks = 4096;
+  if (blocks < 1) blocks = 1;
+  if (bf16)
+    hipLaunchKernelGGL(bag_gather_kernel<uint16_t>, dim3(blocks), dim3(threads),
+                       0, stream, (const uint16_t*)X, bag_idx, bags, T, n, d);
+  else
+    hipLaunchKernelGGL(bag_gather_kernel<float>, dim3(blocks), dim3(threads), 0,
+                       stream, (const float*)X, bag_idx, bags, T, n, d);
+}
+
+void launch_build_forest(const float* bags, const int32_t* feat_sub,
+                         int32_t* feat, float* value, int32_t* right,
+                         int32_t* count, int32_t* ncount,
+                         const float* leaf_lut, uint64_t seed,
+                         int32_t tree_id_offset, int32_t T, int32_t n,
+                         int32_t d, int32_t k, int32_t max_nodes,
+                         int32_t height_limit, size_t lds,
+                         hipStream_t stream) {
+  raise_lds((const void*)build_forest_kernel, lds);
+  hipLaunchKernelGGL(build_forest_kernel, dim3(T), dim3(WAVE), lds, stream,
+                     bags, feat_sub, feat, value, right, count, ncount,
+                     leaf_lut, seed, tree_id_offset, n, d, k, max_nodes,
+                     height_limit);
+}
+
+void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
+                                  int32_t* feat, float* value, int32_t* right,
+                                  int32_t* count, int32_t* ncount,
+                                  int32_t* hidx, float* hw, double* off64,
+                                  const float* leaf_lut, uint64_t seed,
+                                  int32_t tree_id_offset, int32_t T, int32_t n,
+                                  int32_t d, int32_t k, int32_t nnz,
+                                  int32_t max_nodes, int32_t height_limit,
+                                  size_t lds, hipStream_t stream) {
+  raise_lds((const void*)build_extended_forest_kernel, lds);
+  hipLaunchKernelGGL(build_extended_forest_kernel, dim3(T), dim3(WAVE), lds,
+                     stream, bags, feat_sub, feat, value, right, count, ncount,
+                     hidx, hw, off64, leaf_lut, seed, tree_id_offset, n, d, k,
+                     nnz, max_nodes, height_limit);
+}
+
+void launch_score_forest(bool bf16, bool rows_lds, const void* X,
+                         const int4* nodes, const int32_t* ncount, float* out,
+                         int64_t N, int32_t d, int32_t T, int32_t max_nodes,
+                         float fT, float c_norm, int finalize, size_t lds,
+                         int blocks, hipStream_t stream) {
+#define LS(XT, RL)                                                            \
+  do {                                                                        \
+    raise_lds((const void*)score_forest_kernel<XT, RL>, lds);                 \
+    hipLaunchKernelGGL((score_forest_kernel<XT, RL>), dim3(blocks), dim3(256),\
+                       lds, stream, (const XT*)X, nodes, ncount, out, N, d, T,\
+                       max_nodes, fT, c_norm, finalize);                      \
+  } while (0)
+  if (bf16) {
+    if (rows_lds) LS(uint16_t, true); else LS(uint16_t, false);
+  } else {
+    if (rows_lds) LS(float, true); else LS(float, false);
+  }
+#undef LS
+}
+
+void launch_score_extended_forest(bool bf16, bool rows_lds, bool hyper_lds,
+                                  const void* X, const int4* nodes,
+                                  const int32_t* hidx, const float* hw,
+                                  const int32_t* ncount, float* out, int64_t N,
+                                  int32_t d, int32_t T, int32_t max_nodes,
+                                  int32_t nnz, float fT, float c_norm,
+                                  int finalize, size_t lds, int blocks,
+                                  hipStream_t stream) {
+#define LSE(XT, RL, HL)                                                       \
+  do {                                                                        \
+    raise_lds((const void*)score_extended_forest_kernel<XT, RL, HL>, lds);    \
+    hipLaunchKernelGGL((score_extended_forest_kernel<XT, RL, HL>),            \
+                       dim3(blocks), dim3(256), lds, stream, (const XT*)X,    \
+                       nodes, hidx, hw, ncount, out, N, d, T, max_nodes, nnz, \
+                       fT, c_norm, finalize);                                 \
+  } while (0)
+  if (bf16) {
+    if (rows_lds && hyper_lds) LSE(uint16_t, true, true);
+    else if (rows_lds) LSE(uint16_t, true, false);
+    else if (hyper_lds) LSE(uint16_t, false, true);
+    else LSE(uint16_t, false, false);
+  } else {
+    if (rows_lds && hyper_lds) LSE(float, true, true);
+    else if (rows_lds) LSE(float, true, false);
+    else if (hyper_lds) LSE(float, false, true);
+    else LSE(float, false, false);
+  }
+#undef LSE
+}
+
+}  // namespace ifa

@@ -1,0 +1,210 @@
+"""GPU tests (MI355X): bitwise CPU<->GPU parity, bf16 path, end-to-end quality.
+
+All marked @pytest.mark.gpu; they run on the GPU box via
+`python -m pytest tests -m gpu` and are skipped where no GPU exists.
+"""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from isolation_forest_amd import ExtendedIsolationForest, IsolationForest
+from isolation_forest_amd.core import cpu_engine
+from tests.conftest import auroc
+
+
+@pytest.fixture(scope="module")
+def dev():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from isolation_forest_amd.ops import load_extension
+
+    load_extension()  # hard error if missing on a GPU box
+    return torch.device("cuda:0")
+
+
+def make_data(n=5000, d=8, seed=3):
+    rs = np.random.RandomState(seed)
+    return rs.normal(size=(n, d)).astype(np.float32)
+
+
+class TestBagGather:
+    def test_matches_cpu_gather(self, dev):
+        from isolation_forest_amd.ops import load_extension
+
+        ext = load_extension()
+        X = make_data(1000, 8)
+        idx = cpu_engine.sample_bags(1000, 10, 64, seed=1, bootstrap=False)
+        Xt = torch.from_numpy(X).to(dev)
+        bags = ext.bag_gather(Xt, torch.from_numpy(idx).to(dev))
+        expect = X[idx]  # [10, 64, 8]
+        np.testing.assert_array_equal(bags.cpu().numpy(), expect)
+
+    def test_bf16_gather(self, dev):
+        from isolation_forest_amd.ops import load_extension
+
+        ext = load_extension()
+        X = make_data(500, 4)
+        Xb = torch.from_numpy(X).to(dev).to(torch.bfloat16)
+        idx = cpu_engine.sample_bags(500, 5, 32, seed=2, bootstrap=False)
+        bags = ext.bag_gather(Xb, torch.from_numpy(idx).to(dev))
+        expect = Xb.float().cpu().numpy()[idx]
+        np.testing.assert_array_equal(bags.cpu().numpy(), expect)
+
+
+class TestStructuralParity:
+    """The strongest GPU check: GPU-built forests are BIT-IDENTICAL to the
+    CPU oracle's for the same seed."""
+
+    @pytest.mark.parametrize("n,d,k,T", [(256, 8, 8, 16), (128, 16, 5, 8), (1024, 4, 4, 4)])
+    def test_standard_build_identical(self, dev, n, d, k, T):
+        from isolation_forest_amd.ops import gpu_engine
+        from isolation_forest_amd.utils.params import ResolvedParams
+
+        X = make_data(8000, d)
+        bag = cpu_engine.sample_bags(8000, T, n, seed=9, bootstrap=False)
+        fs = cpu_engine.feature_subsets(d, k, T, seed=9)
+        cpu_f = cpu_engine.build_forest(X, bag, fs, 9, n, k, d)
+        rp = ResolvedParams(num_samples=n, num_features=k, total_rows=8000,
+                            total_features=d)
+        gpu_f = gpu_engine.build_forest(torch.from_numpy(X).to(dev), bag, fs, 9, rp)
+        np.testing.assert_array_equal(gpu_f.node_count, cpu_f.node_count)
+        np.testing.assert_array_equal(gpu_f.feature, cpu_f.feature)
+        np.testing.assert_array_equal(gpu_f.right, cpu_f.right)
+        np.testing.assert_array_equal(gpu_f.num_instances, cpu_f.num_instances)
+        np.testing.assert_array_equal(
+            gpu_f.value.view(np.int32), cpu_f.value.view(np.int32)
+        )
+
+    @pytest.mark.parametrize("ext_level", [0, 2, 7])
+    def test_extended_build_identical(self, dev, ext_level):
+        from isolation_forest_amd.ops import gpu_engine
+        from isolation_forest_amd.utils.params import ResolvedParams
+
+        n, d, T = 256, 8, 8
+        X = make_data(6000, d)
+        bag = cpu_engine.sample_bags(6000, T, n, seed=4, bootstrap=False)
+        fs = cpu_engine.feature_subsets(d, d, T, seed=4)
+        cpu_f = cpu_engine.build_extended_forest(X, bag, fs, 4, n, d, d, ext_level)
+        rp = ResolvedParams(num_samples=n, num_features=d, total_rows=6000,
+                            total_features=d, extension_level=ext_level)
+        gpu_f = gpu_engine.build_extended_forest(
+            torch.from_numpy(X).to(dev), bag, fs, 4, rp
+        )
+        np.testing.assert_array_equal(gpu_f.node_count, cpu_f.node_count)
+        np.testing.assert_array_equal(gpu_f.feature, cpu_f.feature)
+        np.testing.assert_array_equal(gpu_f.right, cpu_f.right)
+        np.testing.assert_array_equal(gpu_f.num_instances, cpu_f.num_instances)
+        np.testing.assert_array_equal(gpu_f.hyper_idx, cpu_f.hyper_idx)
+        np.testing.assert_array_equal(
+            gpu_f.hyper_w.view(np.int32), cpu_f.hyper_w.view(np.int32)
+        )
+        np.testing.assert_array_equal(
+            gpu_f.value.view(np.int32), cpu_f.value.view(np.int32)
+        )
+        # offsets are doubles computed in identical order
+        np.testing.assert_array_equal(gpu_f.offset64, cpu_f.offset64)
+
+
+class TestScoringParity:
+    def test_path_sums_bitwise(self, dev):
+        from isolation_forest_amd.ops import gpu_engine
+
+        X = make_data(4000, 8)
+        bag = cpu_engine.sample_bags(4000, 32, 256, seed=2, bootstrap=False)
+        fs = cpu_engine.feature_subsets(8, 8, 32, seed=2)
+        forest = cpu_engine.build_forest(X, bag, fs, 2, 256, 8, 8)
+        cpu_ps = cpu_engine.path_lengths(forest, X)
+
+        model = IsolationForest(numEstimators=32).fit(X[:600])  # shell for cache
+        model.forest = forest
+        model._gpu_forest_cache = {}
+        gpu_ps = gpu_engine.score_forest(
+            model, torch.from_numpy(X).to(dev), finalize=False
+        )
+        np.testing.assert_array_equal(
+            gpu_ps.cpu().numpy().view(np.int32), cpu_ps.view(np.int32)
+        )
+
+    def test_scores_close(self, dev):
+        X = make_data(3000, 8)
+        model = IsolationForest(numEstimators=50, randomSeed=6).fit(X)
+        cpu_scores = model.score(torch.from_numpy(X)).numpy()
+        gpu_scores = model.score(torch.from_numpy(X).to(dev)).cpu().numpy()
+        np.testing.assert_allclose(gpu_scores, cpu_scores, rtol=0, atol=2e-7)
+
+    def test_extended_path_sums_bitwise(self, dev):
+        from isolation_forest_amd.ops import gpu_engine
+
+        X = make_data(2000, 6)
+        bag = cpu_engine.sample_bags(2000, 16, 128, seed=3, bootstrap=False)
+        fs = cpu_engine.feature_subsets(6, 6, 16, seed=3)
+        forest = cpu_engine.build_extended_forest(X, bag, fs, 3, 128, 6, 6, 5)
+        cpu_ps = cpu_engine.path_lengths_extended(forest, X)
+
+        model = ExtendedIsolationForest(numEstimators=16).fit(X[:600])
+        model.forest = forest
+        model._gpu_forest_cache = {}
+        gpu_ps = gpu_engine.score_extended_forest(
+            model, torch.from_numpy(X).to(dev), finalize=False
+        )
+        np.testing.assert_array_equal(
+            gpu_ps.cpu().numpy().view(np.int32), cpu_ps.view(np.int32)
+        )
+
+
+class TestEndToEndGPU:
+    def test_fit_transform_on_device(self, dev):
+        rs = np.random.RandomState(7)
+        d = 8
+        inl = rs.normal(size=(20000, d)).astype(np.float32)
+        out = rs.uniform(-10, 12, size=(400, d)).astype(np.float32)
+        X = np.concatenate([inl, out])
+        y = np.concatenate([np.zeros(20000), np.ones(400)])
+        Xt = torch.from_numpy(X).to(dev)
+        model = IsolationForest(
+            numEstimators=100, contamination=0.02, contaminationError=0.0
+        ).fit(Xt)
+        res = model.transform(Xt)
+        scores = res["outlierScore"].cpu().numpy()
+        assert auroc(y, scores) > 0.9
+        frac = float(res["predictedLabel"].cpu().float().mean())
+        assert frac == pytest.approx(0.02, abs=0.005)
+
+    def test_fit_transform_bf16(self, dev):
+        rs = np.random.RandomState(8)
+        X = rs.normal(size=(30000, 32)).astype(np.float32)
+        Xb = torch.from_numpy(X).to(dev).to(torch.bfloat16)
+        model = IsolationForest(numEstimators=64).fit(Xb)
+        scores = model.transform(Xb)["outlierScore"]
+        assert scores.is_cuda
+        assert float(scores.min()) > 0 and float(scores.max()) < 1
+        # bf16 GPU fit == CPU fit on the bf16-rounded matrix (same seed)
+        Xr = Xb.float().cpu().numpy()
+        cpu_model = IsolationForest(numEstimators=64).fit(Xr)
+        for t in range(0, 64, 16):
+            assert (
+                model.forest.tree_to_string(t) == cpu_model.forest.tree_to_string(t)
+            )
+
+    def test_extended_fit_on_device(self, dev):
+        rs = np.random.RandomState(9)
+        X = torch.from_numpy(rs.normal(size=(10000, 6)).astype(np.float32)).to(dev)
+        model = ExtendedIsolationForest(numEstimators=50).fit(X)
+        scores = model.transform(X)["outlierScore"]
+        assert scores.is_cuda and scores.shape[0] == 10000
+
+    def test_gpu_model_roundtrip_to_cpu(self, dev, tmp_path):
+        X = make_data(3000, 8)
+        Xt = torch.from_numpy(X).to(dev)
+        model = IsolationForest(numEstimators=20, contamination=0.05).fit(Xt)
+        path = str(tmp_path / "gpu_model")
+        model.save(path)
+        from isolation_forest_amd import IsolationForestModel
+
+        loaded = IsolationForestModel.load(path)
+        s_gpu = model.score(Xt).cpu().numpy()
+        s_cpu = loaded.score(torch.from_numpy(X)).numpy()
+        np.testing.assert_allclose(s_cpu, s_gpu, rtol=0, atol=2e-7)
