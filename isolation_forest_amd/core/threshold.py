@@ -92,7 +92,12 @@ def compute_threshold(
         hist = _allreduce(hist.double(), comm)
         cum = torch.cumsum(hist, dim=0)
         target = k - rank_below_lo
-        bin_idx = int(torch.searchsorted(cum, torch.tensor(float(target), dtype=torch.float64)).item())
+        bin_idx = int(
+            torch.searchsorted(
+                cum,
+                torch.tensor(float(target), dtype=torch.float64, device=cum.device),
+            ).item()
+        )
         bin_idx = min(bin_idx, _NBINS - 1)
         in_bin = int(hist[bin_idx].item())
         below = int(cum[bin_idx - 1].item()) if bin_idx > 0 else 0

@@ -28,6 +28,10 @@ ext = CUDAExtension(
             "-O3",
             "-std=c++17",
             "--offload-arch=gfx950",
+            # determinism contract: no fma contraction — device doubles must
+            # be bit-identical to the numpy oracle (utils/det_math.py); the
+            # hot float32 paths use explicit __f{add,mul,div}_rn already.
+            "-ffp-contract=off",
         ],
     },
 )
