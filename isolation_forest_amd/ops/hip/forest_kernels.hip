@@ -343,8 +343,12 @@ __global__ void __launch_bounds__(WAVE) build_extended_forest_kernel(
       }
       continue;
     }
-    const float norm = __fsqrt_rn(acc);
-    for (int j = lane; j < nnz; j += WAVE) wf[j] = __fdiv_rn(wf[j], norm);
+    // sqrt/divide through double: __fsqrt_rn/__fdiv_rn are NOT correctly
+    // rounded on this toolchain (1 ulp), numpy's f32 ops are; f64-then-round
+    // is provably identical to correctly-rounded f32 (53 >= 2*24+2).
+    const float norm = (float)sqrt((double)acc);
+    for (int j = lane; j < nnz; j += WAVE)
+      wf[j] = (float)((double)wf[j] / (double)norm);
     __syncthreads();
 
     // ---- intercepts per sorted coordinate + offset (float64) ----
@@ -494,7 +498,7 @@ __global__ void __launch_bounds__(256) score_forest_kernel(
     }
     if (my_row < N) {
       if (finalize) {
-        const float mean32 = __fdiv_rn(path_sum, fT);
+        const float mean32 = (float)((double)path_sum / (double)fT);
         const double ratio = (double)mean32 / (double)c_norm;
         out[my_row] = (float)exp2(-ratio);
       } else {
@@ -591,7 +595,7 @@ __global__ void __launch_bounds__(256) score_extended_forest_kernel(
     }
     if (my_row < N) {
       if (finalize) {
-        const float mean32 = __fdiv_rn(path_sum, fT);
+        const float mean32 = (float)((double)path_sum / (double)fT);
         const double ratio = (double)mean32 / (double)c_norm;
         out[my_row] = (float)exp2(-ratio);
       } else {
