@@ -91,6 +91,7 @@ def build_extended_forest(X: torch.Tensor, bag_idx: np.ndarray,
     c_np = count.cpu().numpy().astype(np.int64)
     nc_np = ncount.cpu().numpy()
     live = _mask_padding(f_np, r_np, c_np, nc_np, ExtendedForest.PAD)
+    v_np = np.where(live, v_np, 0.0).astype(np.float32)
     hidx_np = hidx.cpu().numpy()
     hw_np = hw.cpu().numpy()
     off64_np = off64.cpu().numpy()
