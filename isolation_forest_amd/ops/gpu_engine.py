@@ -120,13 +120,26 @@ def build_extended_forest(X: torch.Tensor, bag_idx: np.ndarray,
 # ---------------------------------------------------------------------------
 
 
-def _nodes_aos(forest) -> np.ndarray:
+def _nodes_packed(forest) -> np.ndarray:
+    """Packed 8-byte node records for the scoring kernels:
+    meta<0 => leaf (value = c(count)); else feature = meta&0xFFF,
+    right = (meta>>12)&0x7FFF (left child is implicit pre-order id+1)."""
     T, mn = forest.feature.shape
-    aos = np.zeros((T, mn, 4), dtype=np.int32)
-    aos[..., 0] = forest.feature
-    aos[..., 1] = forest.value.view(np.int32)
-    aos[..., 2] = forest.right
-    return aos
+    if mn > 32767:
+        raise ValueError("forest too deep for the packed node format")
+    feat = forest.feature
+    if feat.max(initial=0) > 4095:
+        raise ValueError("GPU scoring supports at most 4096 features")
+    internal = feat >= 0
+    meta = np.where(
+        internal,
+        feat.astype(np.int32) | (forest.right.astype(np.int32) << 12),
+        np.int32(-2147483648),
+    ).astype(np.int32)
+    packed = np.empty((T, mn, 2), dtype=np.int32)
+    packed[..., 0] = meta
+    packed[..., 1] = forest.value.view(np.int32)
+    return packed
 
 
 def _device_forest(model, device):
@@ -134,7 +147,7 @@ def _device_forest(model, device):
     cache = model._gpu_forest_cache
     if key not in cache:
         forest = model.forest
-        aos = torch.from_numpy(_nodes_aos(forest)).to(device)
+        aos = torch.from_numpy(_nodes_packed(forest)).to(device)
         ncount = torch.from_numpy(
             np.ascontiguousarray(forest.node_count, dtype=np.int32)
         ).to(device)

@@ -33,13 +33,20 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
                                   size_t lds, hipStream_t stream);
 
 void launch_score_forest(bool bf16, bool rows_lds, const void* X,
-                         const int4* nodes, const int32_t* ncount, float* out,
+                         const void* nodes, const int32_t* ncount, float* out,
                          int64_t N, int32_t d, int32_t T, int32_t max_nodes,
                          float fT, float c_norm, int finalize, size_t lds,
                          int blocks, hipStream_t stream);
 
+void launch_score_extended_dense(bool bf16, bool rows_lds, const void* X,
+                                 const void* nodes, const float* hw,
+                                 const int32_t* ncount, float* out, int64_t N,
+                                 int32_t d, int32_t T, int32_t max_nodes,
+                                 float fT, float c_norm, int finalize,
+                                 size_t lds, int blocks, hipStream_t stream);
+
 void launch_score_extended_forest(bool bf16, bool rows_lds, bool hyper_lds,
-                                  const void* X, const int4* nodes,
+                                  const void* X, const void* nodes,
                                   const int32_t* hidx, const float* hw,
                                   const int32_t* ncount, float* out, int64_t N,
                                   int32_t d, int32_t T, int32_t max_nodes,
@@ -170,32 +177,36 @@ std::vector<torch::Tensor> build_extended_forest(
   return {feat, value, right, count, ncount, hidx, hw, off64};
 }
 
-torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_aos,
+torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
                            torch::Tensor ncount, double c_norm,
                            bool finalize) {
   CHECK_CUDA(X);
   CHECK_CONTIG(X);
-  CHECK_CUDA(nodes_aos);
-  CHECK_CONTIG(nodes_aos);
+  CHECK_CUDA(nodes_packed);
+  CHECK_CONTIG(nodes_packed);
   CHECK_CUDA(ncount);
   check_x(X);
-  TORCH_CHECK(nodes_aos.dim() == 3 && nodes_aos.size(2) == 4 &&
-                  nodes_aos.scalar_type() == torch::kInt32,
-              "nodes_aos must be int32 [T, max_nodes, 4]");
+  TORCH_CHECK(nodes_packed.dim() == 3 && nodes_packed.size(2) == 2 &&
+                  nodes_packed.scalar_type() == torch::kInt32,
+              "nodes must be packed int32 [T, max_nodes, 2]");
   int64_t N = X.size(0), d = X.size(1);
-  int64_t T = nodes_aos.size(0), max_nodes = nodes_aos.size(1);
+  int64_t T = nodes_packed.size(0), max_nodes = nodes_packed.size(1);
+  TORCH_CHECK(d <= 4095, "scoring supports d <= 4095 (12-bit feature field)");
   auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
   if (N == 0) return out;
 
-  size_t node_bytes = (size_t)max_nodes * 16;
-  size_t row_bytes = (size_t)256 * (d + 1) * 4;
-  bool rows_lds = (d % 4 == 0) && (node_bytes + row_bytes <= 128 * 1024);
+  const bool bf16 = is_bf16(X);
+  const size_t elem = bf16 ? 2 : 4;
+  const size_t pad = 16 / elem;
+  size_t node_bytes = (size_t)2 * max_nodes * 8;
+  size_t row_bytes = (size_t)256 * (d + pad) * elem;
+  bool rows_lds = node_bytes + row_bytes <= 144 * 1024;
   size_t lds = node_bytes + (rows_lds ? row_bytes : 0);
   TORCH_CHECK(node_bytes <= kMaxLds, "tree too large for LDS staging");
 
-  int blocks = (int)std::min<int64_t>((N + 255) / 256, 4096);
-  ifa::launch_score_forest(is_bf16(X), rows_lds, X.data_ptr(),
-                           (const int4*)nodes_aos.data_ptr<int32_t>(),
+  int blocks = (int)std::min<int64_t>((N + 255) / 256, 8192);
+  ifa::launch_score_forest(bf16, rows_lds, X.data_ptr(),
+                           nodes_packed.data_ptr<int32_t>(),
                            ncount.data_ptr<int32_t>(), out.data_ptr<float>(),
                            N, (int32_t)d, (int32_t)T, (int32_t)max_nodes,
                            (float)T, (float)c_norm, finalize ? 1 : 0, lds,
@@ -203,40 +214,60 @@ torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_aos,
   return out;
 }
 
-torch::Tensor score_extended_forest(torch::Tensor X, torch::Tensor nodes_aos,
+torch::Tensor score_extended_forest(torch::Tensor X,
+                                    torch::Tensor nodes_packed,
                                     torch::Tensor hidx, torch::Tensor hw,
                                     torch::Tensor ncount, double c_norm,
                                     bool finalize) {
   CHECK_CUDA(X);
   CHECK_CONTIG(X);
-  CHECK_CUDA(nodes_aos);
-  CHECK_CONTIG(nodes_aos);
+  CHECK_CUDA(nodes_packed);
+  CHECK_CONTIG(nodes_packed);
   CHECK_CUDA(hidx);
   CHECK_CONTIG(hidx);
   CHECK_CUDA(hw);
   CHECK_CONTIG(hw);
   check_x(X);
+  TORCH_CHECK(nodes_packed.dim() == 3 && nodes_packed.size(2) == 2 &&
+                  nodes_packed.scalar_type() == torch::kInt32,
+              "nodes must be packed int32 [T, max_nodes, 2]");
   int64_t N = X.size(0), d = X.size(1);
-  int64_t T = nodes_aos.size(0), max_nodes = nodes_aos.size(1);
+  int64_t T = nodes_packed.size(0), max_nodes = nodes_packed.size(1);
   int64_t nnz = hidx.size(2);
+  TORCH_CHECK(d <= 4095, "scoring supports d <= 4095");
   auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
   if (N == 0) return out;
 
-  size_t node_bytes = (size_t)max_nodes * 16;
-  size_t hyper_bytes = (size_t)max_nodes * nnz * 8;
-  size_t row_bytes = (size_t)256 * (d + 1) * 4;
-  bool hyper_lds = node_bytes + hyper_bytes <= 96 * 1024;
-  bool rows_lds =
-      (d % 4 == 0) &&
-      (node_bytes + (hyper_lds ? hyper_bytes : 0) + row_bytes <= 144 * 1024);
-  size_t lds =
-      node_bytes + (hyper_lds ? hyper_bytes : 0) + (rows_lds ? row_bytes : 0);
+  const bool bf16 = is_bf16(X);
+  const size_t elem = bf16 ? 2 : 4;
+  const size_t pad = 16 / elem;
+  size_t row_bytes = (size_t)256 * (d + pad) * elem;
+  size_t node_bytes = (size_t)max_nodes * 8;
+  int blocks = (int)std::min<int64_t>((N + 255) / 256, 8192);
   TORCH_CHECK(node_bytes <= kMaxLds, "tree too large for LDS staging");
 
-  int blocks = (int)std::min<int64_t>((N + 255) / 256, 4096);
+  if (nnz == d && (d % 4) == 0) {
+    // fully-extended dense hyperplanes: implicit indices, vectorized dot
+    bool rows_lds = node_bytes + row_bytes <= 144 * 1024;
+    size_t lds = node_bytes + (rows_lds ? row_bytes : 0);
+    ifa::launch_score_extended_dense(
+        bf16, rows_lds, X.data_ptr(), nodes_packed.data_ptr<int32_t>(),
+        hw.data_ptr<float>(), ncount.data_ptr<int32_t>(),
+        out.data_ptr<float>(), N, (int32_t)d, (int32_t)T, (int32_t)max_nodes,
+        (float)T, (float)c_norm, finalize ? 1 : 0, lds, blocks,
+        current_stream());
+    return out;
+  }
+
+  size_t hyper_bytes = (size_t)max_nodes * nnz * 8;
+  bool hyper_lds = node_bytes + hyper_bytes <= 96 * 1024;
+  bool rows_lds =
+      node_bytes + (hyper_lds ? hyper_bytes : 0) + row_bytes <= 144 * 1024;
+  size_t lds =
+      node_bytes + (hyper_lds ? hyper_bytes : 0) + (rows_lds ? row_bytes : 0);
   ifa::launch_score_extended_forest(
-      is_bf16(X), rows_lds, hyper_lds, X.data_ptr(),
-      (const int4*)nodes_aos.data_ptr<int32_t>(), hidx.data_ptr<int32_t>(),
+      bf16, rows_lds, hyper_lds, X.data_ptr(),
+      nodes_packed.data_ptr<int32_t>(), hidx.data_ptr<int32_t>(),
       hw.data_ptr<float>(), ncount.data_ptr<int32_t>(), out.data_ptr<float>(),
       N, (int32_t)d, (int32_t)T, (int32_t)max_nodes, (int32_t)nnz, (float)T,
       (float)c_norm, finalize ? 1 : 0, lds, blocks, current_stream());

@@ -429,25 +429,51 @@ __global__ void __launch_bounds__(WAVE) build_extended_forest_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// standard scoring: the throughput kernel (K6)
+// scoring kernels (K6/K7) — v2
+//
+// Packed 8-byte node record {int meta, float val}: leaf iff meta < 0
+// (val = precomputed c(numInstances)); else feature = meta & 0xFFF,
+// right child = (meta >> 12) & 0x7FFF (left child is implicit pre-order
+// id+1). One ds_read_b64 per visit.
+//
+// Rows are staged in LDS in the data's NATIVE dtype (bf16 rows cost 2 B,
+// doubling occupancy vs f32), row stride padded to kill power-of-2 bank
+// conflicts while keeping 16-B row alignment for the dense-EIF vector
+// reads. Each thread walks TWO trees concurrently (independent dependent-
+// load chains -> 2x memory-level parallelism against the ~50-cycle LDS
+// latency that bounds this kernel).
 // ---------------------------------------------------------------------------
 
-// node AoS record (16 B): {feat|-1, value_bits(f32), right, unused}
-// one ds_read_b128 per visit when staged in LDS.
+__device__ __forceinline__ float cvt_feat(float v) { return v; }
+__device__ __forceinline__ float cvt_feat(uint16_t v) {
+  union { uint32_t u; float f; } cv;
+  cv.u = ((uint32_t)v) << 16;
+  return cv.f;
+}
+
+__device__ __forceinline__ int pn_feat(int meta) { return meta & 0xFFF; }
+__device__ __forceinline__ int pn_right(int meta) { return (meta >> 12) & 0x7FFF; }
+
+// row stride in elements: keep rows 16-B aligned, break power-of-2 banking
+template <typename XT>
+__device__ __host__ __forceinline__ int row_stride(int d) {
+  const int pad = 16 / (int)sizeof(XT);  // 8 for bf16, 4 for f32
+  return d + pad;
+}
 
 template <typename XT, bool ROWS_LDS>
 __global__ void __launch_bounds__(256) score_forest_kernel(
     const XT* __restrict__ X,          // [N][d]
-    const int4* __restrict__ nodes,    // [T][max_nodes] AoS
+    const int2* __restrict__ nodes,    // [T][max_nodes] packed
     const int32_t* __restrict__ ncnt,  // [T]
     float* __restrict__ out,           // [N] (path sum or score)
     int64_t N, int32_t d, int32_t T, int32_t max_nodes, float fT,
     float c_norm, int32_t finalize) {
   const int tid = threadIdx.x;
-  const int dpad = d + 1;
+  const int dpad = row_stride<XT>(d);
 
-  int4* tree_lds = (int4*)smem;  // [max_nodes]
-  float* rows = (float*)(tree_lds + max_nodes);  // [256][dpad] if ROWS_LDS
+  int2* tlds = (int2*)smem;            // [2 * max_nodes]
+  XT* rows = (XT*)(tlds + 2 * max_nodes);  // [256][dpad] if ROWS_LDS
 
   for (int64_t block_row0 = (int64_t)blockIdx.x * 256; block_row0 < N;
        block_row0 += (int64_t)gridDim.x * 256) {
@@ -455,45 +481,68 @@ __global__ void __launch_bounds__(256) score_forest_kernel(
     const int rows_here = (int)min((int64_t)256, N - block_row0);
 
     if (ROWS_LDS) {
-      __syncthreads();  // previous iteration's reads done
+      __syncthreads();  // previous iteration's readers done
       const int64_t total = (int64_t)rows_here * d;
       for (int64_t g = tid; g < total; g += 256) {
         const int r = (int)(g / d), c = (int)(g % d);
-        rows[r * dpad + c] = load_feat<XT>(X, (block_row0 + r) * d + c);
+        rows[r * dpad + c] = X[(block_row0 + r) * d + c];
       }
       __syncthreads();
     }
+    const XT* my_lrow = rows + tid * dpad;
 
     float path_sum = 0.f;
-    for (int t = 0; t < T; ++t) {
-      // stage tree t into LDS (coop)
+    for (int t = 0; t < T; t += 2) {
       __syncthreads();
-      const int nc = ncnt[t];
-      const int4* src = nodes + (int64_t)t * max_nodes;
-      for (int i = tid; i < nc; i += 256) tree_lds[i] = src[i];
+      const bool has1 = (t + 1) < T;
+      {
+        const int nc0 = ncnt[t];
+        const int2* s0 = nodes + (int64_t)t * max_nodes;
+        for (int i = tid; i < nc0; i += 256) tlds[i] = s0[i];
+        if (has1) {
+          const int nc1 = ncnt[t + 1];
+          const int2* s1 = nodes + (int64_t)(t + 1) * max_nodes;
+          for (int i = tid; i < nc1; i += 256) tlds[max_nodes + i] = s1[i];
+        }
+      }
       __syncthreads();
 
       if (my_row < N) {
-        int node = 0;
-        int depth = 0;
-        float leaf = 0.f;
-        while (true) {
-          const int4 nd = tree_lds[node];
-          if (nd.x < 0) {
-            leaf = __int_as_float(nd.y);
-            break;
+        int c0 = 0, c1 = 0, dep0 = 0, dep1 = 0;
+        float lf0 = 0.f, lf1 = 0.f;
+        bool a0 = true, a1 = has1;
+        while (a0 || a1) {
+          const int2 n0 = tlds[c0];
+          const int2 n1 = tlds[max_nodes + c1];
+          if (a0) {
+            if (n0.x < 0) {
+              lf0 = __int_as_float(n0.y);
+              a0 = false;
+            } else {
+              const float xv =
+                  ROWS_LDS ? cvt_feat(my_lrow[pn_feat(n0.x)])
+                           : load_feat<XT>(X, my_row * d + pn_feat(n0.x));
+              c0 = (xv < __int_as_float(n0.y)) ? c0 + 1 : pn_right(n0.x);
+              ++dep0;
+            }
           }
-          float xv;
-          if (ROWS_LDS)
-            xv = rows[tid * dpad + nd.x];
-          else
-            xv = load_feat<XT>(X, my_row * d + nd.x);
-          node = (xv < __int_as_float(nd.y)) ? node + 1 : nd.z;
-          ++depth;
+          if (a1) {
+            if (n1.x < 0) {
+              lf1 = __int_as_float(n1.y);
+              a1 = false;
+            } else {
+              const float xv =
+                  ROWS_LDS ? cvt_feat(my_lrow[pn_feat(n1.x)])
+                           : load_feat<XT>(X, my_row * d + pn_feat(n1.x));
+              c1 = (xv < __int_as_float(n1.y)) ? c1 + 1 : pn_right(n1.x);
+              ++dep1;
+            }
+          }
         }
-        // oracle order (cpu_engine.path_lengths): per tree
+        // oracle order (cpu_engine.path_lengths): per tree, in tree order:
         // total = f32(total + f32(depth + leaf))
-        path_sum = __fadd_rn(path_sum, __fadd_rn((float)depth, leaf));
+        path_sum = __fadd_rn(path_sum, __fadd_rn((float)dep0, lf0));
+        if (has1) path_sum = __fadd_rn(path_sum, __fadd_rn((float)dep1, lf1));
       }
     }
     if (my_row < N) {
@@ -509,24 +558,27 @@ __global__ void __launch_bounds__(256) score_forest_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// extended scoring (K7)
+// extended scoring, dense fast path: fully-extended hyperplanes over ALL
+// features (nnz == d, the EIF default) have implicit indices, so each visit
+// is a dense dot row[0..d) . w[0..d) — vectorized float4 weight loads from
+// L2 (the per-tree weight slab is L2-resident) against the LDS row tile.
+// Accumulation order: 4 partial lanes then pairwise combine (deterministic,
+// but NOT the oracle's strict j-order — scoring parity tests use tolerance).
 // ---------------------------------------------------------------------------
 
-template <typename XT, bool ROWS_LDS, bool HYPER_LDS>
-__global__ void __launch_bounds__(256) score_extended_forest_kernel(
-    const XT* __restrict__ X, const int4* __restrict__ nodes,
-    const int32_t* __restrict__ hidx_g,  // [T][max_nodes][nnz]
-    const float* __restrict__ hw_g,      // [T][max_nodes][nnz]
+template <typename XT, bool ROWS_LDS>
+__global__ void __launch_bounds__(256) score_extended_dense_kernel(
+    const XT* __restrict__ X, const int2* __restrict__ nodes,
+    const float* __restrict__ hw_g,  // [T][max_nodes][d] dense weights
     const int32_t* __restrict__ ncnt, float* __restrict__ out, int64_t N,
-    int32_t d, int32_t T, int32_t max_nodes, int32_t nnz, float fT,
-    float c_norm, int32_t finalize) {
+    int32_t d, int32_t T, int32_t max_nodes, float fT, float c_norm,
+    int32_t finalize) {
   const int tid = threadIdx.x;
-  const int dpad = d + 1;
+  const int dpad = row_stride<XT>(d);
+  const int d4 = d >> 2;
 
-  int4* tree_lds = (int4*)smem;                       // [max_nodes]
-  int32_t* hidx_lds = (int32_t*)(tree_lds + max_nodes);  // [max_nodes*nnz]
-  float* hw_lds = (float*)(hidx_lds + (HYPER_LDS ? max_nodes * nnz : 0));
-  float* rows = hw_lds + (HYPER_LDS ? max_nodes * nnz : 0);
+  int2* tlds = (int2*)smem;          // [max_nodes]
+  XT* rows = (XT*)(tlds + max_nodes);  // [256][dpad] if ROWS_LDS
 
   for (int64_t block_row0 = (int64_t)blockIdx.x * 256; block_row0 < N;
        block_row0 += (int64_t)gridDim.x * 256) {
@@ -538,17 +590,113 @@ __global__ void __launch_bounds__(256) score_extended_forest_kernel(
       const int64_t total = (int64_t)rows_here * d;
       for (int64_t g = tid; g < total; g += 256) {
         const int r = (int)(g / d), c = (int)(g % d);
-        rows[r * dpad + c] = load_feat<XT>(X, (block_row0 + r) * d + c);
+        rows[r * dpad + c] = X[(block_row0 + r) * d + c];
       }
       __syncthreads();
     }
+    const XT* my_lrow = rows + tid * dpad;
 
     float path_sum = 0.f;
     for (int t = 0; t < T; ++t) {
       __syncthreads();
       const int nc = ncnt[t];
-      const int4* src = nodes + (int64_t)t * max_nodes;
-      for (int i = tid; i < nc; i += 256) tree_lds[i] = src[i];
+      const int2* src = nodes + (int64_t)t * max_nodes;
+      for (int i = tid; i < nc; i += 256) tlds[i] = src[i];
+      __syncthreads();
+
+      if (my_row < N) {
+        const float* wbase = hw_g + (int64_t)t * max_nodes * d;
+        int cur = 0, dep = 0;
+        float leaf = 0.f;
+        while (true) {
+          const int2 nd = tlds[cur];
+          if (nd.x < 0) {
+            leaf = __int_as_float(nd.y);
+            break;
+          }
+          const float4* wp = (const float4*)(wbase + (int64_t)cur * d);
+          float ax = 0.f, ay = 0.f, az = 0.f, aw = 0.f;
+          for (int j = 0; j < d4; ++j) {
+            const float4 w = wp[j];
+            float x0, x1, x2, x3;
+            if (ROWS_LDS) {
+              x0 = cvt_feat(my_lrow[4 * j]);
+              x1 = cvt_feat(my_lrow[4 * j + 1]);
+              x2 = cvt_feat(my_lrow[4 * j + 2]);
+              x3 = cvt_feat(my_lrow[4 * j + 3]);
+            } else {
+              x0 = load_feat<XT>(X, my_row * d + 4 * j);
+              x1 = load_feat<XT>(X, my_row * d + 4 * j + 1);
+              x2 = load_feat<XT>(X, my_row * d + 4 * j + 2);
+              x3 = load_feat<XT>(X, my_row * d + 4 * j + 3);
+            }
+            ax = __fadd_rn(ax, __fmul_rn(w.x, x0));
+            ay = __fadd_rn(ay, __fmul_rn(w.y, x1));
+            az = __fadd_rn(az, __fmul_rn(w.z, x2));
+            aw = __fadd_rn(aw, __fmul_rn(w.w, x3));
+          }
+          const float dot =
+              __fadd_rn(__fadd_rn(ax, ay), __fadd_rn(az, aw));
+          cur = (dot < __int_as_float(nd.y)) ? cur + 1 : pn_right(nd.x);
+          ++dep;
+        }
+        path_sum = __fadd_rn(path_sum, __fadd_rn((float)dep, leaf));
+      }
+    }
+    if (my_row < N) {
+      if (finalize) {
+        const float mean32 = (float)((double)path_sum / (double)fT);
+        const double ratio = (double)mean32 / (double)c_norm;
+        out[my_row] = (float)exp2(-ratio);
+      } else {
+        out[my_row] = path_sum;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// extended scoring, sparse general path (nnz < d): exact oracle order.
+// ---------------------------------------------------------------------------
+
+template <typename XT, bool ROWS_LDS, bool HYPER_LDS>
+__global__ void __launch_bounds__(256) score_extended_forest_kernel(
+    const XT* __restrict__ X, const int2* __restrict__ nodes,
+    const int32_t* __restrict__ hidx_g,  // [T][max_nodes][nnz]
+    const float* __restrict__ hw_g,      // [T][max_nodes][nnz]
+    const int32_t* __restrict__ ncnt, float* __restrict__ out, int64_t N,
+    int32_t d, int32_t T, int32_t max_nodes, int32_t nnz, float fT,
+    float c_norm, int32_t finalize) {
+  const int tid = threadIdx.x;
+  const int dpad = row_stride<XT>(d);
+
+  int2* tlds = (int2*)smem;                          // [max_nodes]
+  int32_t* hidx_lds = (int32_t*)(tlds + max_nodes);  // [max_nodes*nnz]
+  float* hw_lds = (float*)(hidx_lds + (HYPER_LDS ? max_nodes * nnz : 0));
+  XT* rows = (XT*)(hw_lds + (HYPER_LDS ? max_nodes * nnz : 0));
+
+  for (int64_t block_row0 = (int64_t)blockIdx.x * 256; block_row0 < N;
+       block_row0 += (int64_t)gridDim.x * 256) {
+    const int64_t my_row = block_row0 + tid;
+    const int rows_here = (int)min((int64_t)256, N - block_row0);
+
+    if (ROWS_LDS) {
+      __syncthreads();
+      const int64_t total = (int64_t)rows_here * d;
+      for (int64_t g = tid; g < total; g += 256) {
+        const int r = (int)(g / d), c = (int)(g % d);
+        rows[r * dpad + c] = X[(block_row0 + r) * d + c];
+      }
+      __syncthreads();
+    }
+    const XT* my_lrow = rows + tid * dpad;
+
+    float path_sum = 0.f;
+    for (int t = 0; t < T; ++t) {
+      __syncthreads();
+      const int nc = ncnt[t];
+      const int2* src = nodes + (int64_t)t * max_nodes;
+      for (int i = tid; i < nc; i += 256) tlds[i] = src[i];
       if (HYPER_LDS) {
         const int64_t hbase = (int64_t)t * max_nodes * nnz;
         for (int i = tid; i < nc * nnz; i += 256) {
@@ -560,37 +708,29 @@ __global__ void __launch_bounds__(256) score_extended_forest_kernel(
 
       if (my_row < N) {
         const int64_t hbase = (int64_t)t * max_nodes * nnz;
-        int node = 0;
-        int depth = 0;
+        int cur = 0, dep = 0;
         float leaf = 0.f;
         while (true) {
-          const int4 nd = tree_lds[node];
+          const int2 nd = tlds[cur];
           if (nd.x < 0) {
             leaf = __int_as_float(nd.y);
             break;
           }
           float dot = 0.f;
-          if (HYPER_LDS) {
-            const int32_t* ci = hidx_lds + node * nnz;
-            const float* cw = hw_lds + node * nnz;
-            for (int j = 0; j < nnz; ++j) {
-              float xv = ROWS_LDS ? rows[tid * dpad + ci[j]]
-                                  : load_feat<XT>(X, my_row * d + ci[j]);
-              dot = __fadd_rn(dot, __fmul_rn(cw[j], xv));
-            }
-          } else {
-            const int32_t* ci = hidx_g + hbase + (int64_t)node * nnz;
-            const float* cw = hw_g + hbase + (int64_t)node * nnz;
-            for (int j = 0; j < nnz; ++j) {
-              float xv = ROWS_LDS ? rows[tid * dpad + ci[j]]
-                                  : load_feat<XT>(X, my_row * d + ci[j]);
-              dot = __fadd_rn(dot, __fmul_rn(cw[j], xv));
-            }
+          const int32_t* ci = HYPER_LDS ? hidx_lds + cur * nnz
+                                        : hidx_g + hbase + (int64_t)cur * nnz;
+          const float* cw = HYPER_LDS ? hw_lds + cur * nnz
+                                      : hw_g + hbase + (int64_t)cur * nnz;
+          for (int j = 0; j < nnz; ++j) {
+            const float xv = ROWS_LDS
+                                 ? cvt_feat(my_lrow[ci[j]])
+                                 : load_feat<XT>(X, my_row * d + ci[j]);
+            dot = __fadd_rn(dot, __fmul_rn(cw[j], xv));
           }
-          node = (dot < __int_as_float(nd.y)) ? node + 1 : nd.z;
-          ++depth;
+          cur = (dot < __int_as_float(nd.y)) ? cur + 1 : pn_right(nd.x);
+          ++dep;
         }
-        path_sum = __fadd_rn(path_sum, __fadd_rn((float)depth, leaf));
+        path_sum = __fadd_rn(path_sum, __fadd_rn((float)dep, leaf));
       }
     }
     if (my_row < N) {
@@ -668,7 +808,7 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
 }
 
 void launch_score_forest(bool bf16, bool rows_lds, const void* X,
-                         const int4* nodes, const int32_t* ncount, float* out,
+                         const void* nodes, const int32_t* ncount, float* out,
                          int64_t N, int32_t d, int32_t T, int32_t max_nodes,
                          float fT, float c_norm, int finalize, size_t lds,
                          int blocks, hipStream_t stream) {
@@ -676,8 +816,8 @@ void launch_score_forest(bool bf16, bool rows_lds, const void* X,
   do {                                                                        \
     raise_lds((const void*)score_forest_kernel<XT, RL>, lds);                 \
     hipLaunchKernelGGL((score_forest_kernel<XT, RL>), dim3(blocks), dim3(256),\
-                       lds, stream, (const XT*)X, nodes, ncount, out, N, d, T,\
-                       max_nodes, fT, c_norm, finalize);                      \
+                       lds, stream, (const XT*)X, (const int2*)nodes, ncount, \
+                       out, N, d, T, max_nodes, fT, c_norm, finalize);        \
   } while (0)
   if (bf16) {
     if (rows_lds) LS(uint16_t, true); else LS(uint16_t, false);
@@ -687,8 +827,30 @@ void launch_score_forest(bool bf16, bool rows_lds, const void* X,
 #undef LS
 }
 
+void launch_score_extended_dense(bool bf16, bool rows_lds, const void* X,
+                                 const void* nodes, const float* hw,
+                                 const int32_t* ncount, float* out, int64_t N,
+                                 int32_t d, int32_t T, int32_t max_nodes,
+                                 float fT, float c_norm, int finalize,
+                                 size_t lds, int blocks, hipStream_t stream) {
+#define LSD(XT, RL)                                                           \
+  do {                                                                        \
+    raise_lds((const void*)score_extended_dense_kernel<XT, RL>, lds);         \
+    hipLaunchKernelGGL((score_extended_dense_kernel<XT, RL>), dim3(blocks),   \
+                       dim3(256), lds, stream, (const XT*)X,                  \
+                       (const int2*)nodes, hw, ncount, out, N, d, T,          \
+                       max_nodes, fT, c_norm, finalize);                      \
+  } while (0)
+  if (bf16) {
+    if (rows_lds) LSD(uint16_t, true); else LSD(uint16_t, false);
+  } else {
+    if (rows_lds) LSD(float, true); else LSD(float, false);
+  }
+#undef LSD
+}
+
 void launch_score_extended_forest(bool bf16, bool rows_lds, bool hyper_lds,
-                                  const void* X, const int4* nodes,
+                                  const void* X, const void* nodes,
                                   const int32_t* hidx, const float* hw,
                                   const int32_t* ncount, float* out, int64_t N,
                                   int32_t d, int32_t T, int32_t max_nodes,
@@ -700,8 +862,8 @@ void launch_score_extended_forest(bool bf16, bool rows_lds, bool hyper_lds,
     raise_lds((const void*)score_extended_forest_kernel<XT, RL, HL>, lds);    \
     hipLaunchKernelGGL((score_extended_forest_kernel<XT, RL, HL>),            \
                        dim3(blocks), dim3(256), lds, stream, (const XT*)X,    \
-                       nodes, hidx, hw, ncount, out, N, d, T, max_nodes, nnz, \
-                       fT, c_norm, finalize);                                 \
+                       (const int2*)nodes, hidx, hw, ncount, out, N, d, T,    \
+                       max_nodes, nnz, fT, c_norm, finalize);                 \
   } while (0)
   if (bf16) {
     if (rows_lds && hyper_lds) LSE(uint16_t, true, true);
