@@ -34,14 +34,16 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
 
 void launch_score_forest(bool bf16, bool rows_lds, const void* X,
                          const void* nodes, const int32_t* ncount, float* out,
-                         int64_t N, int32_t d, int32_t T, int32_t max_nodes,
-                         float fT, float c_norm, int finalize, size_t lds,
-                         int blocks, hipStream_t stream);
+                         int64_t N, int32_t d, int32_t dpad, int32_t T,
+                         int32_t max_nodes, float fT, float c_norm,
+                         int finalize, size_t lds, int blocks,
+                         hipStream_t stream);
 
-void launch_score_extended_dense(bool bf16, bool rows_lds, const void* X,
-                                 const void* nodes, const float* hw,
-                                 const int32_t* ncount, float* out, int64_t N,
-                                 int32_t d, int32_t T, int32_t max_nodes,
+void launch_score_extended_dense(bool bf16, bool rows_lds, bool wlds,
+                                 const void* X, const void* nodes,
+                                 const float* hw, const int32_t* ncount,
+                                 float* out, int64_t N, int32_t d,
+                                 int32_t dpad, int32_t T, int32_t max_nodes,
                                  float fT, float c_norm, int finalize,
                                  size_t lds, int blocks, hipStream_t stream);
 
@@ -197,10 +199,17 @@ torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
 
   const bool bf16 = is_bf16(X);
   const size_t elem = bf16 ? 2 : 4;
-  const size_t pad = 16 / elem;
-  size_t node_bytes = (size_t)2 * max_nodes * 8;
-  size_t row_bytes = (size_t)256 * (d + pad) * elem;
-  bool rows_lds = node_bytes + row_bytes <= 144 * 1024;
+  // scalar-read kernel: odd-WORD row stride so lane bases hit all 32 banks
+  int64_t dpad;
+  if (bf16) {
+    int64_t pad = (2 - (d % 4) + 4) % 4;  // (d+pad) % 4 == 2
+    dpad = d + pad;
+  } else {
+    dpad = (d % 2 == 0) ? d + 1 : d;  // odd word count
+  }
+  size_t node_bytes = (size_t)4 * max_nodes * 8;  // ILP=4 staged trees
+  size_t row_bytes = (size_t)256 * dpad * elem;
+  bool rows_lds = node_bytes + row_bytes <= 150 * 1024;
   size_t lds = node_bytes + (rows_lds ? row_bytes : 0);
   TORCH_CHECK(node_bytes <= kMaxLds, "tree too large for LDS staging");
 
@@ -208,9 +217,9 @@ torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
   ifa::launch_score_forest(bf16, rows_lds, X.data_ptr(),
                            nodes_packed.data_ptr<int32_t>(),
                            ncount.data_ptr<int32_t>(), out.data_ptr<float>(),
-                           N, (int32_t)d, (int32_t)T, (int32_t)max_nodes,
-                           (float)T, (float)c_norm, finalize ? 1 : 0, lds,
-                           blocks, current_stream());
+                           N, (int32_t)d, (int32_t)dpad, (int32_t)T,
+                           (int32_t)max_nodes, (float)T, (float)c_norm,
+                           finalize ? 1 : 0, lds, blocks, current_stream());
   return out;
 }
 
@@ -247,15 +256,22 @@ torch::Tensor score_extended_forest(torch::Tensor X,
   TORCH_CHECK(node_bytes <= kMaxLds, "tree too large for LDS staging");
 
   if (nnz == d && (d % 4) == 0) {
-    // fully-extended dense hyperplanes: implicit indices, vectorized dot
-    bool rows_lds = node_bytes + row_bytes <= 144 * 1024;
-    size_t lds = node_bytes + (rows_lds ? row_bytes : 0);
+    // fully-extended dense hyperplanes: implicit indices, vectorized dot.
+    // 8-B-aligned row stride for the uint2/float4 row reads.
+    int64_t pad4 = (4 - (d % 4)) % 4;
+    int64_t dpad = d + (pad4 == 0 ? 4 : pad4);
+    size_t drow_bytes = (size_t)256 * dpad * elem;
+    size_t w_bytes = (size_t)max_nodes * d * 4;
+    bool wlds = node_bytes + w_bytes + drow_bytes <= 152 * 1024;
+    bool rows_lds =
+        node_bytes + (wlds ? w_bytes : 0) + drow_bytes <= 152 * 1024;
+    size_t lds = node_bytes + (wlds ? w_bytes : 0) + (rows_lds ? drow_bytes : 0);
     ifa::launch_score_extended_dense(
-        bf16, rows_lds, X.data_ptr(), nodes_packed.data_ptr<int32_t>(),
+        bf16, rows_lds, wlds, X.data_ptr(), nodes_packed.data_ptr<int32_t>(),
         hw.data_ptr<float>(), ncount.data_ptr<int32_t>(),
-        out.data_ptr<float>(), N, (int32_t)d, (int32_t)T, (int32_t)max_nodes,
-        (float)T, (float)c_norm, finalize ? 1 : 0, lds, blocks,
-        current_stream());
+        out.data_ptr<float>(), N, (int32_t)d, (int32_t)dpad, (int32_t)T,
+        (int32_t)max_nodes, (float)T, (float)c_norm, finalize ? 1 : 0, lds,
+        blocks, current_stream());
     return out;
   }
 

@@ -461,19 +461,22 @@ __device__ __host__ __forceinline__ int row_stride(int d) {
   return d + pad;
 }
 
-template <typename XT, bool ROWS_LDS>
-__global__ void __launch_bounds__(256) score_forest_kernel(
+// rows are staged at a runtime element stride `dpad` chosen by the host:
+// an ODD word count for the scalar-read kernels (all 32 banks hit —
+// Guideline 4) and a 16-B-aligned stride for the vector-read dense kernel.
+
+template <typename XT, bool ROWS_LDS, int ILP>
+__global__ void __launch_bounds__(256) score_forest_kernel_t(
     const XT* __restrict__ X,          // [N][d]
     const int2* __restrict__ nodes,    // [T][max_nodes] packed
     const int32_t* __restrict__ ncnt,  // [T]
     float* __restrict__ out,           // [N] (path sum or score)
-    int64_t N, int32_t d, int32_t T, int32_t max_nodes, float fT,
-    float c_norm, int32_t finalize) {
+    int64_t N, int32_t d, int32_t dpad, int32_t T, int32_t max_nodes,
+    float fT, float c_norm, int32_t finalize) {
   const int tid = threadIdx.x;
-  const int dpad = row_stride<XT>(d);
 
-  int2* tlds = (int2*)smem;            // [2 * max_nodes]
-  XT* rows = (XT*)(tlds + 2 * max_nodes);  // [256][dpad] if ROWS_LDS
+  int2* tlds = (int2*)smem;                // [ILP * max_nodes]
+  XT* rows = (XT*)(tlds + ILP * max_nodes);  // [256][dpad] if ROWS_LDS
 
   for (int64_t block_row0 = (int64_t)blockIdx.x * 256; block_row0 < N;
        block_row0 += (int64_t)gridDim.x * 256) {
@@ -492,57 +495,62 @@ __global__ void __launch_bounds__(256) score_forest_kernel(
     const XT* my_lrow = rows + tid * dpad;
 
     float path_sum = 0.f;
-    for (int t = 0; t < T; t += 2) {
+    for (int t = 0; t < T; t += ILP) {
       __syncthreads();
-      const bool has1 = (t + 1) < T;
-      {
-        const int nc0 = ncnt[t];
-        const int2* s0 = nodes + (int64_t)t * max_nodes;
-        for (int i = tid; i < nc0; i += 256) tlds[i] = s0[i];
-        if (has1) {
-          const int nc1 = ncnt[t + 1];
-          const int2* s1 = nodes + (int64_t)(t + 1) * max_nodes;
-          for (int i = tid; i < nc1; i += 256) tlds[max_nodes + i] = s1[i];
+      const int live = min(ILP, T - t);
+#pragma unroll
+      for (int s = 0; s < ILP; ++s) {
+        if (s < live) {
+          const int ncs = ncnt[t + s];
+          const int2* ss = nodes + (int64_t)(t + s) * max_nodes;
+          for (int i = tid; i < ncs; i += 256) tlds[s * max_nodes + i] = ss[i];
         }
       }
       __syncthreads();
 
       if (my_row < N) {
-        int c0 = 0, c1 = 0, dep0 = 0, dep1 = 0;
-        float lf0 = 0.f, lf1 = 0.f;
-        bool a0 = true, a1 = has1;
-        while (a0 || a1) {
-          const int2 n0 = tlds[c0];
-          const int2 n1 = tlds[max_nodes + c1];
-          if (a0) {
-            if (n0.x < 0) {
-              lf0 = __int_as_float(n0.y);
-              a0 = false;
-            } else {
-              const float xv =
-                  ROWS_LDS ? cvt_feat(my_lrow[pn_feat(n0.x)])
-                           : load_feat<XT>(X, my_row * d + pn_feat(n0.x));
-              c0 = (xv < __int_as_float(n0.y)) ? c0 + 1 : pn_right(n0.x);
-              ++dep0;
-            }
-          }
-          if (a1) {
-            if (n1.x < 0) {
-              lf1 = __int_as_float(n1.y);
-              a1 = false;
-            } else {
-              const float xv =
-                  ROWS_LDS ? cvt_feat(my_lrow[pn_feat(n1.x)])
-                           : load_feat<XT>(X, my_row * d + pn_feat(n1.x));
-              c1 = (xv < __int_as_float(n1.y)) ? c1 + 1 : pn_right(n1.x);
-              ++dep1;
+        int cur[ILP], dep[ILP];
+        float lf[ILP];
+        bool act[ILP];
+        bool any = false;
+#pragma unroll
+        for (int s = 0; s < ILP; ++s) {
+          cur[s] = 0;
+          dep[s] = 0;
+          lf[s] = 0.f;
+          act[s] = s < live;
+          any |= act[s];
+        }
+        while (any) {
+          int2 nd[ILP];
+#pragma unroll
+          for (int s = 0; s < ILP; ++s) nd[s] = tlds[s * max_nodes + cur[s]];
+          any = false;
+#pragma unroll
+          for (int s = 0; s < ILP; ++s) {
+            if (act[s]) {
+              if (nd[s].x < 0) {
+                lf[s] = __int_as_float(nd[s].y);
+                act[s] = false;
+              } else {
+                const float xv =
+                    ROWS_LDS
+                        ? cvt_feat(my_lrow[pn_feat(nd[s].x)])
+                        : load_feat<XT>(X, my_row * d + pn_feat(nd[s].x));
+                cur[s] =
+                    (xv < __int_as_float(nd[s].y)) ? cur[s] + 1 : pn_right(nd[s].x);
+                ++dep[s];
+                any = true;
+              }
             }
           }
         }
         // oracle order (cpu_engine.path_lengths): per tree, in tree order:
         // total = f32(total + f32(depth + leaf))
-        path_sum = __fadd_rn(path_sum, __fadd_rn((float)dep0, lf0));
-        if (has1) path_sum = __fadd_rn(path_sum, __fadd_rn((float)dep1, lf1));
+#pragma unroll
+        for (int s = 0; s < ILP; ++s)
+          if (s < live)
+            path_sum = __fadd_rn(path_sum, __fadd_rn((float)dep[s], lf[s]));
       }
     }
     if (my_row < N) {
@@ -558,27 +566,44 @@ __global__ void __launch_bounds__(256) score_forest_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// extended scoring, dense fast path: fully-extended hyperplanes over ALL
-// features (nnz == d, the EIF default) have implicit indices, so each visit
-// is a dense dot row[0..d) . w[0..d) — vectorized float4 weight loads from
-// L2 (the per-tree weight slab is L2-resident) against the LDS row tile.
-// Accumulation order: 4 partial lanes then pairwise combine (deterministic,
-// but NOT the oracle's strict j-order — scoring parity tests use tolerance).
+// extended scoring, dense fast path (nnz == d): per visit a dense dot
+// row[0..d) . w[0..d). WLDS stages the live tree's weight matrix in LDS
+// (L2 could not feed the ~17 TB/s the global-weight variant demanded);
+// rows are read as 8-byte vectors (4 bf16 or 2 f32 at a time).
+// Accumulation: 4 partial sums, pairwise combine (deterministic, not the
+// oracle's j-order — scoring parity tests use tolerance).
 // ---------------------------------------------------------------------------
 
-template <typename XT, bool ROWS_LDS>
+__device__ __forceinline__ void row4(const uint16_t* p, int j4, float& x0,
+                                     float& x1, float& x2, float& x3) {
+  const uint2 v = *(const uint2*)(p + 4 * j4);  // 8 B: 4 bf16
+  union { uint32_t u; float f; } a, b, c, e;
+  a.u = (v.x & 0xFFFFu) << 16;
+  b.u = (v.x & 0xFFFF0000u);
+  c.u = (v.y & 0xFFFFu) << 16;
+  e.u = (v.y & 0xFFFF0000u);
+  x0 = a.f; x1 = b.f; x2 = c.f; x3 = e.f;
+}
+
+__device__ __forceinline__ void row4(const float* p, int j4, float& x0,
+                                     float& x1, float& x2, float& x3) {
+  const float4 v = *(const float4*)(p + 4 * j4);
+  x0 = v.x; x1 = v.y; x2 = v.z; x3 = v.w;
+}
+
+template <typename XT, bool ROWS_LDS, bool WLDS>
 __global__ void __launch_bounds__(256) score_extended_dense_kernel(
     const XT* __restrict__ X, const int2* __restrict__ nodes,
     const float* __restrict__ hw_g,  // [T][max_nodes][d] dense weights
     const int32_t* __restrict__ ncnt, float* __restrict__ out, int64_t N,
-    int32_t d, int32_t T, int32_t max_nodes, float fT, float c_norm,
-    int32_t finalize) {
+    int32_t d, int32_t dpad, int32_t T, int32_t max_nodes, float fT,
+    float c_norm, int32_t finalize) {
   const int tid = threadIdx.x;
-  const int dpad = row_stride<XT>(d);
   const int d4 = d >> 2;
 
-  int2* tlds = (int2*)smem;          // [max_nodes]
-  XT* rows = (XT*)(tlds + max_nodes);  // [256][dpad] if ROWS_LDS
+  int2* tlds = (int2*)smem;                  // [max_nodes]
+  float* wlds = (float*)(tlds + max_nodes);  // [max_nodes * d] if WLDS
+  XT* rows = (XT*)(wlds + (WLDS ? max_nodes * d : 0));
 
   for (int64_t block_row0 = (int64_t)blockIdx.x * 256; block_row0 < N;
        block_row0 += (int64_t)gridDim.x * 256) {
@@ -602,10 +627,17 @@ __global__ void __launch_bounds__(256) score_extended_dense_kernel(
       const int nc = ncnt[t];
       const int2* src = nodes + (int64_t)t * max_nodes;
       for (int i = tid; i < nc; i += 256) tlds[i] = src[i];
+      if (WLDS) {
+        const float4* wsrc = (const float4*)(hw_g + (int64_t)t * max_nodes * d);
+        float4* wdst = (float4*)wlds;
+        const int n4 = nc * d4;
+        for (int i = tid; i < n4; i += 256) wdst[i] = wsrc[i];
+      }
       __syncthreads();
 
       if (my_row < N) {
-        const float* wbase = hw_g + (int64_t)t * max_nodes * d;
+        const float* wbase =
+            WLDS ? wlds : hw_g + (int64_t)t * max_nodes * d;
         int cur = 0, dep = 0;
         float leaf = 0.f;
         while (true) {
@@ -620,10 +652,7 @@ __global__ void __launch_bounds__(256) score_extended_dense_kernel(
             const float4 w = wp[j];
             float x0, x1, x2, x3;
             if (ROWS_LDS) {
-              x0 = cvt_feat(my_lrow[4 * j]);
-              x1 = cvt_feat(my_lrow[4 * j + 1]);
-              x2 = cvt_feat(my_lrow[4 * j + 2]);
-              x3 = cvt_feat(my_lrow[4 * j + 3]);
+              row4(my_lrow, j, x0, x1, x2, x3);
             } else {
               x0 = load_feat<XT>(X, my_row * d + 4 * j);
               x1 = load_feat<XT>(X, my_row * d + 4 * j + 1);
@@ -809,15 +838,17 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
 
 void launch_score_forest(bool bf16, bool rows_lds, const void* X,
                          const void* nodes, const int32_t* ncount, float* out,
-                         int64_t N, int32_t d, int32_t T, int32_t max_nodes,
-                         float fT, float c_norm, int finalize, size_t lds,
-                         int blocks, hipStream_t stream) {
+                         int64_t N, int32_t d, int32_t dpad, int32_t T,
+                         int32_t max_nodes, float fT, float c_norm,
+                         int finalize, size_t lds, int blocks,
+                         hipStream_t stream) {
 #define LS(XT, RL)                                                            \
   do {                                                                        \
-    raise_lds((const void*)score_forest_kernel<XT, RL>, lds);                 \
-    hipLaunchKernelGGL((score_forest_kernel<XT, RL>), dim3(blocks), dim3(256),\
-                       lds, stream, (const XT*)X, (const int2*)nodes, ncount, \
-                       out, N, d, T, max_nodes, fT, c_norm, finalize);        \
+    raise_lds((const void*)score_forest_kernel_t<XT, RL, 4>, lds);            \
+    hipLaunchKernelGGL((score_forest_kernel_t<XT, RL, 4>), dim3(blocks),      \
+                       dim3(256), lds, stream, (const XT*)X,                  \
+                       (const int2*)nodes, ncount, out, N, d, dpad, T,        \
+                       max_nodes, fT, c_norm, finalize);                      \
   } while (0)
   if (bf16) {
     if (rows_lds) LS(uint16_t, true); else LS(uint16_t, false);
@@ -827,24 +858,31 @@ void launch_score_forest(bool bf16, bool rows_lds, const void* X,
 #undef LS
 }
 
-void launch_score_extended_dense(bool bf16, bool rows_lds, const void* X,
-                                 const void* nodes, const float* hw,
-                                 const int32_t* ncount, float* out, int64_t N,
-                                 int32_t d, int32_t T, int32_t max_nodes,
+void launch_score_extended_dense(bool bf16, bool rows_lds, bool wlds,
+                                 const void* X, const void* nodes,
+                                 const float* hw, const int32_t* ncount,
+                                 float* out, int64_t N, int32_t d,
+                                 int32_t dpad, int32_t T, int32_t max_nodes,
                                  float fT, float c_norm, int finalize,
                                  size_t lds, int blocks, hipStream_t stream) {
-#define LSD(XT, RL)                                                           \
+#define LSD(XT, RL, WL)                                                       \
   do {                                                                        \
-    raise_lds((const void*)score_extended_dense_kernel<XT, RL>, lds);         \
-    hipLaunchKernelGGL((score_extended_dense_kernel<XT, RL>), dim3(blocks),   \
-                       dim3(256), lds, stream, (const XT*)X,                  \
-                       (const int2*)nodes, hw, ncount, out, N, d, T,          \
+    raise_lds((const void*)score_extended_dense_kernel<XT, RL, WL>, lds);     \
+    hipLaunchKernelGGL((score_extended_dense_kernel<XT, RL, WL>),             \
+                       dim3(blocks), dim3(256), lds, stream, (const XT*)X,    \
+                       (const int2*)nodes, hw, ncount, out, N, d, dpad, T,    \
                        max_nodes, fT, c_norm, finalize);                      \
   } while (0)
   if (bf16) {
-    if (rows_lds) LSD(uint16_t, true); else LSD(uint16_t, false);
+    if (rows_lds && wlds) LSD(uint16_t, true, true);
+    else if (rows_lds) LSD(uint16_t, true, false);
+    else if (wlds) LSD(uint16_t, false, true);
+    else LSD(uint16_t, false, false);
   } else {
-    if (rows_lds) LSD(float, true); else LSD(float, false);
+    if (rows_lds && wlds) LSD(float, true, true);
+    else if (rows_lds) LSD(float, true, false);
+    else if (wlds) LSD(float, false, true);
+    else LSD(float, false, false);
   }
 #undef LSD
 }
