@@ -120,8 +120,113 @@ def build_extended_forest(X: torch.Tensor, bag_idx: np.ndarray,
 # ---------------------------------------------------------------------------
 
 
+_BF16_KEY_TABLES = None
+
+
+def _bf16_key_tables():
+    """(values, keys) of every producible bf16 KEY, value-ascending.
+
+    The key transform is the device's key16(): order-preserving u16 with
+    +-0 collapsed to 0x8000 and NaN at 0xFFFF. 'Producible' = what the
+    staging transform can emit for non-NaN input: finite and +-inf keys,
+    excluding 0x7FFF (-0 maps to +0's key)."""
+    global _BF16_KEY_TABLES
+    if _BF16_KEY_TABLES is None:
+        keys = np.arange(65536, dtype=np.uint32)
+        pos = keys >= 0x8000
+        bits = np.where(pos, keys - 0x8000,
+                        0x8000 + (0x7FFF - keys)).astype(np.uint32)
+        vals = (bits << np.uint32(16)).view(np.float32)
+        ok = np.isfinite(vals) | np.isinf(vals)
+        ok[0x7FFF] = False  # -0's raw key: never staged
+        _BF16_KEY_TABLES = (vals[ok].copy(), keys[ok].astype(np.uint32))
+    return _BF16_KEY_TABLES
+
+
+def _key32(vals: np.ndarray) -> np.ndarray:
+    """Order-preserving u32 key of f32 values (device key32())."""
+    b = np.ascontiguousarray(vals, dtype=np.float32).view(np.uint32)
+    m = b & np.uint32(0x7FFFFFFF)
+    k = np.where(b & np.uint32(0x80000000),
+                 np.uint32(0x7FFFFFFF) - m,
+                 np.uint32(0x80000000) + m).astype(np.uint32)
+    k[m == 0] = np.uint32(0x80000000)
+    k[m > np.uint32(0x7F800000)] = np.uint32(0xFFFFFFFF)
+    return k
+
+
+def _bf16_threshold_keys(split_vals: np.ndarray) -> np.ndarray:
+    """Smallest producible bf16 key whose value is >= the f32 split value,
+    shifted into the high 16 bits (kernel compares widened keys)."""
+    vals, keys = _bf16_key_tables()
+    idx = np.searchsorted(vals, split_vals.astype(np.float32), side="left")
+    return keys[idx].astype(np.uint32) << np.uint32(16)
+
+
+def _node_depths(feature: np.ndarray, right: np.ndarray) -> np.ndarray:
+    """Per-node depth via one pre-order sweep (parents precede children)."""
+    T, mn = feature.shape
+    depth = np.zeros((T, mn), dtype=np.int32)
+    rows = np.arange(T)
+    for i in range(mn - 1):
+        internal = feature[:, i] >= 0
+        if not internal.any():
+            continue
+        d1 = depth[:, i] + 1
+        depth[internal, i + 1] = d1[internal]
+        ri = rows[internal]
+        depth[ri, right[internal, i]] = d1[internal]
+    return depth
+
+
+def _nodes_packed_v4(forest, d_sentinel: int, bf16: bool):
+    """v4 packed nodes for score_forest_v4 (see forest_kernels.hip):
+    w0 = feat | right<<12; internal w1 = integer key threshold; leaf:
+    feat = d_sentinel, right = own id (self-loop), w1 = f32(depth + c(m)).
+    Trees are padded to a multiple of 4 with single-leaf value-0 dummies.
+    Returns (packed int32 [Tpad, mn, 2], ncount int32 [Tpad])."""
+    T, mn = forest.feature.shape
+    if mn > 32767:
+        raise ValueError("forest too deep for the packed node format")
+    feat = forest.feature
+    if feat.max(initial=0) > 4094 or d_sentinel > 4094:
+        raise ValueError("GPU scoring supports at most 4095 features")
+    right = forest.right
+    val = forest.value
+    internal = feat >= 0
+    leaf = feat == Forest.LEAF
+    depth = _node_depths(feat, right)
+    ids = np.broadcast_to(np.arange(mn, dtype=np.int32)[None, :], (T, mn))
+    w0 = np.where(
+        internal,
+        feat.astype(np.int32) | (right.astype(np.int32) << 12),
+        np.int32(d_sentinel) | (ids << 12),
+    ).astype(np.int32)
+    w1 = np.zeros((T, mn), dtype=np.uint32)
+    if internal.any():
+        s = val[internal].astype(np.float32)
+        w1[internal] = _bf16_threshold_keys(s) if bf16 else _key32(s)
+    leafval = (depth.astype(np.float32) + val.astype(np.float32))
+    w1[leaf] = leafval[leaf].astype(np.float32).view(np.uint32)
+    pad = (-T) % 4
+    if pad:
+        w0p = np.empty((pad, mn), dtype=np.int32)
+        w0p[:] = np.int32(d_sentinel) | (ids[0] << 12)
+        w0 = np.concatenate([w0, w0p])
+        w1 = np.concatenate([w1, np.zeros((pad, mn), dtype=np.uint32)])
+    packed = np.empty((T + pad, mn, 2), dtype=np.int32)
+    packed[..., 0] = w0
+    packed[..., 1] = w1.view(np.int32)
+    ncount = np.concatenate(
+        [forest.node_count.astype(np.int32), np.ones(pad, dtype=np.int32)]
+    )
+    live = internal | leaf
+    max_depth = int(depth[live].max()) if live.any() else 0
+    return packed, ncount, max(max_depth, 1)
+
+
 def _nodes_packed(forest) -> np.ndarray:
-    """Packed 8-byte node records for the scoring kernels:
+    """Packed 8-byte node records for the EIF scoring kernels:
     meta<0 => leaf (value = c(count)); else feature = meta&0xFFF,
     right = (meta>>12)&0x7FFF (left child is implicit pre-order id+1)."""
     T, mn = forest.feature.shape
@@ -142,23 +247,32 @@ def _nodes_packed(forest) -> np.ndarray:
     return packed
 
 
-def _device_forest(model, device):
-    key = (str(device), id(model.forest))
+def _device_forest(model, device, v4_key=None):
+    """Cached device copy of the packed forest. v4_key = (d, bf16) selects
+    the standard-scoring v4 packing; None = the EIF packing."""
+    key = (str(device), id(model.forest), v4_key)
     cache = model._gpu_forest_cache
     if key not in cache:
         forest = model.forest
-        aos = torch.from_numpy(_nodes_packed(forest)).to(device)
-        ncount = torch.from_numpy(
-            np.ascontiguousarray(forest.node_count, dtype=np.int32)
-        ).to(device)
         extra = {}
-        if isinstance(forest, ExtendedForest):
-            extra["hidx"] = torch.from_numpy(
-                np.ascontiguousarray(forest.hyper_idx)
+        if v4_key is not None:
+            d, bf16 = v4_key
+            packed, ncount_np, max_depth = _nodes_packed_v4(forest, d, bf16)
+            aos = torch.from_numpy(packed).to(device)
+            ncount = torch.from_numpy(ncount_np).to(device)
+            extra["height"] = max_depth
+        else:
+            aos = torch.from_numpy(_nodes_packed(forest)).to(device)
+            ncount = torch.from_numpy(
+                np.ascontiguousarray(forest.node_count, dtype=np.int32)
             ).to(device)
-            extra["hw"] = torch.from_numpy(
-                np.ascontiguousarray(forest.hyper_w)
-            ).to(device)
+            if isinstance(forest, ExtendedForest):
+                extra["hidx"] = torch.from_numpy(
+                    np.ascontiguousarray(forest.hyper_idx)
+                ).to(device)
+                extra["hw"] = torch.from_numpy(
+                    np.ascontiguousarray(forest.hyper_w)
+                ).to(device)
         cache.clear()  # one cached device copy per model is enough
         cache[key] = (aos, ncount, extra)
     return cache[key]
@@ -167,9 +281,14 @@ def _device_forest(model, device):
 def score_forest(model, X: torch.Tensor, finalize: bool = True) -> torch.Tensor:
     ext = load_extension()
     forest = model.forest
-    aos, ncount, _ = _device_forest(model, X.device)
+    d = int(X.shape[1])
+    bf16 = X.dtype == torch.bfloat16
+    aos, ncount, extra = _device_forest(model, X.device, v4_key=(d, bf16))
     c = float(avg_path_length(forest.num_samples))
-    return ext.score_forest(X.contiguous(), aos, ncount, c, finalize)
+    return ext.score_forest(
+        X.contiguous(), aos, ncount, forest.num_trees, extra["height"], c,
+        finalize,
+    )
 
 
 def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torch.Tensor:

@@ -32,11 +32,11 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
                                   int32_t max_nodes, int32_t height_limit,
                                   size_t lds, hipStream_t stream);
 
-void launch_score_forest(bool bf16, bool rows_lds, const void* X,
+void launch_score_forest(bool bf16, int rpt, bool rows_lds, const void* X,
                          const void* nodes, const int32_t* ncount, float* out,
-                         int64_t N, int32_t d, int32_t dpad, int32_t T,
-                         int32_t max_nodes, float fT, float c_norm,
-                         int finalize, size_t lds, int blocks,
+                         int64_t N, int32_t d, int32_t dpad, int32_t Tpad,
+                         int32_t max_nodes, int32_t height_limit, float fT,
+                         float c_norm, int finalize, size_t lds, int blocks,
                          hipStream_t stream);
 
 void launch_score_extended_dense(bool bf16, bool rows_lds, bool wlds,
@@ -180,7 +180,8 @@ std::vector<torch::Tensor> build_extended_forest(
 }
 
 torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
-                           torch::Tensor ncount, double c_norm,
+                           torch::Tensor ncount, int64_t num_trees,
+                           int64_t height_limit, double c_norm,
                            bool finalize) {
   CHECK_CUDA(X);
   CHECK_CONTIG(X);
@@ -190,36 +191,50 @@ torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
   check_x(X);
   TORCH_CHECK(nodes_packed.dim() == 3 && nodes_packed.size(2) == 2 &&
                   nodes_packed.scalar_type() == torch::kInt32,
-              "nodes must be packed int32 [T, max_nodes, 2]");
+              "nodes must be packed int32 [Tpad, max_nodes, 2]");
   int64_t N = X.size(0), d = X.size(1);
-  int64_t T = nodes_packed.size(0), max_nodes = nodes_packed.size(1);
-  TORCH_CHECK(d <= 4095, "scoring supports d <= 4095 (12-bit feature field)");
+  int64_t Tpad = nodes_packed.size(0), max_nodes = nodes_packed.size(1);
+  TORCH_CHECK(Tpad % 4 == 0, "packed tree count must be padded to 4");
+  TORCH_CHECK(d <= 4094, "scoring supports d <= 4094 (12-bit feature field "
+                         "plus the leaf sentinel column)");
   auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
   if (N == 0) return out;
 
   const bool bf16 = is_bf16(X);
   const size_t elem = bf16 ? 2 : 4;
-  // scalar-read kernel: odd-WORD row stride so lane bases hit all 32 banks
-  int64_t dpad;
+  // odd-WORD row stride (all 32 LDS banks hit) with room for the sentinel
+  // column at index d
+  int64_t dpad = d + 1;
   if (bf16) {
-    int64_t pad = (2 - (d % 4) + 4) % 4;  // (d+pad) % 4 == 2
-    dpad = d + pad;
+    while (dpad % 4 != 2) ++dpad;  // dpad u16 elems -> dpad/2 words, odd
   } else {
-    dpad = (d % 2 == 0) ? d + 1 : d;  // odd word count
+    while (dpad % 2 != 1) ++dpad;  // odd word count
   }
-  size_t node_bytes = (size_t)4 * max_nodes * 8;  // ILP=4 staged trees
-  size_t row_bytes = (size_t)256 * dpad * elem;
-  bool rows_lds = node_bytes + row_bytes <= 150 * 1024;
-  size_t lds = node_bytes + (rows_lds ? row_bytes : 0);
+  const size_t node_bytes = (size_t)4 * max_nodes * 8;  // ILP=4 staged trees
   TORCH_CHECK(node_bytes <= kMaxLds, "tree too large for LDS staging");
-
-  int blocks = (int)std::min<int64_t>((N + 255) / 256, 8192);
-  ifa::launch_score_forest(bf16, rows_lds, X.data_ptr(),
+  // prefer 2 rows/thread when 3 blocks/CU still fit, else 1, else global X
+  int rpt = 2;
+  bool rows_lds = true;
+  size_t lds;
+  for (;;) {
+    size_t row_bytes = (size_t)rpt * 256 * dpad * elem;
+    lds = node_bytes + row_bytes;
+    if (lds * 3 <= kMaxLds || (rpt == 1 && lds <= 150 * 1024)) break;
+    if (rpt == 2) { rpt = 1; continue; }
+    rows_lds = false;
+    lds = node_bytes;
+    break;
+  }
+  int64_t rows_per_block = (int64_t)(rows_lds ? rpt : 1) * 256;
+  int blocks = (int)std::min<int64_t>(
+      (N + rows_per_block - 1) / rows_per_block, 8192);
+  ifa::launch_score_forest(bf16, rpt, rows_lds, X.data_ptr(),
                            nodes_packed.data_ptr<int32_t>(),
                            ncount.data_ptr<int32_t>(), out.data_ptr<float>(),
-                           N, (int32_t)d, (int32_t)dpad, (int32_t)T,
-                           (int32_t)max_nodes, (float)T, (float)c_norm,
-                           finalize ? 1 : 0, lds, blocks, current_stream());
+                           N, (int32_t)d, (int32_t)dpad, (int32_t)Tpad,
+                           (int32_t)max_nodes, (int32_t)height_limit,
+                           (float)num_trees, (float)c_norm, finalize ? 1 : 0,
+                           lds, blocks, current_stream());
   return out;
 }
 

@@ -429,19 +429,25 @@ __global__ void __launch_bounds__(WAVE) build_extended_forest_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// scoring kernels (K6/K7) — v2
+// standard scoring kernel (K6/K8) — v4
 //
-// Packed 8-byte node record {int meta, float val}: leaf iff meta < 0
-// (val = precomputed c(numInstances)); else feature = meta & 0xFFF,
-// right child = (meta >> 12) & 0x7FFF (left child is implicit pre-order
-// id+1). One ds_read_b64 per visit.
+// Node record is 8 bytes {uint32 w0, uint32 w1}:
+//   w0 = feature(12 bits) | right_child(15 bits) << 12
+//   internal: w1 = integer KEY threshold (order-preserving transform of the
+//             f32 split value; bf16 keys live in the HIGH 16 bits)
+//   leaf:     feature = d (a sentinel column staged as all-ones), right =
+//             OWN id (self-loop), w1 = f32 bits of (depth + c(numInstances))
 //
-// Rows are staged in LDS in the data's NATIVE dtype (bf16 rows cost 2 B,
-// doubling occupancy vs f32), row stride padded to kill power-of-2 bank
-// conflicts while keeping 16-B row alignment for the dense-EIF vector
-// reads. Each thread walks TWO trees concurrently (independent dependent-
-// load chains -> 2x memory-level parallelism against the ~50-cycle LDS
-// latency that bounds this kernel).
+// Rows are staged in LDS as order-preserving integer keys (u16 for bf16
+// input, u32 for f32), so each visit is: ds_read_b64 node, ds_read LDS key,
+// one unsigned compare, one select — no float convert, no leaf branch, no
+// depth counter. Leaves self-loop (sentinel key 0xFF..F compares false
+// against every threshold), so the walk is a FIXED height_limit-trip loop
+// with zero per-chain bookkeeping; the leaf value (depth folded in, as the
+// reference's ONNX converter does: isolation_forest_converter.py:343-373)
+// is read once after the loop. Each thread walks RPT rows x 4 trees = 4*RPT
+// independent dependent-load chains. Bitwise-identical to
+// cpu_engine.path_lengths (tree-order f32 accumulation).
 // ---------------------------------------------------------------------------
 
 __device__ __forceinline__ float cvt_feat(float v) { return v; }
@@ -454,112 +460,156 @@ __device__ __forceinline__ float cvt_feat(uint16_t v) {
 __device__ __forceinline__ int pn_feat(int meta) { return meta & 0xFFF; }
 __device__ __forceinline__ int pn_right(int meta) { return (meta >> 12) & 0x7FFF; }
 
-// row stride in elements: keep rows 16-B aligned, break power-of-2 banking
+// order-preserving key transforms (match gpu_engine._key16/_key32):
+// +-0 collapse to the +0 key, NaN maps to the max key (so `x < t` is false:
+// NaN routes right, same as the f32 compare in the oracle).
+__device__ __forceinline__ uint32_t key16(uint32_t b) {
+  const uint32_t m = b & 0x7FFFu;
+  uint32_t k = (b & 0x8000u) ? (0x7FFFu - m) : (0x8000u + m);
+  if (m == 0) k = 0x8000u;
+  if (m > 0x7F80u) k = 0xFFFFu;
+  return k;
+}
+__device__ __forceinline__ uint32_t key32(uint32_t b) {
+  const uint32_t m = b & 0x7FFFFFFFu;
+  uint32_t k = (b & 0x80000000u) ? (0x7FFFFFFFu - m) : (0x80000000u + m);
+  if (m == 0) k = 0x80000000u;
+  if (m > 0x7F800000u) k = 0xFFFFFFFFu;
+  return k;
+}
+
+// KT = uint16_t (bf16 input) or uint32_t (f32 input)
+template <typename KT>
+__device__ __forceinline__ KT key_of_bits(KT b);
+template <>
+__device__ __forceinline__ uint16_t key_of_bits<uint16_t>(uint16_t b) {
+  return (uint16_t)key16((uint32_t)b);
+}
+template <>
+__device__ __forceinline__ uint32_t key_of_bits<uint32_t>(uint32_t b) {
+  return key32(b);
+}
+
+// widen a stored key for the compare: bf16 keys sit in the high 16 bits with
+// low bits saturated so the sentinel (0xFFFF) widens to 0xFFFFFFFF and
+// compares false against every w1 (including leaf f32 payloads).
+__device__ __forceinline__ uint32_t widen_key(uint16_t k) {
+  return ((uint32_t)k << 16) | 0xFFFFu;
+}
+__device__ __forceinline__ uint32_t widen_key(uint32_t k) { return k; }
+
+// row stride in elements for the extended kernels: keep rows 16-B aligned
 template <typename XT>
 __device__ __host__ __forceinline__ int row_stride(int d) {
   const int pad = 16 / (int)sizeof(XT);  // 8 for bf16, 4 for f32
   return d + pad;
 }
 
-// rows are staged at a runtime element stride `dpad` chosen by the host:
-// an ODD word count for the scalar-read kernels (all 32 banks hit —
-// Guideline 4) and a 16-B-aligned stride for the vector-read dense kernel.
+#define V4_ILP 4
 
-template <typename XT, bool ROWS_LDS, int ILP>
-__global__ void __launch_bounds__(256) score_forest_kernel_t(
-    const XT* __restrict__ X,          // [N][d]
-    const int2* __restrict__ nodes,    // [T][max_nodes] packed
-    const int32_t* __restrict__ ncnt,  // [T]
+template <typename KT, int RPT, bool ROWS_LDS>
+__global__ void __launch_bounds__(256) score_forest_v4(
+    const KT* __restrict__ X,          // raw bf16/f32 bits [N][d]
+    const int2* __restrict__ nodes,    // [Tpad][max_nodes] packed v4
+    const int32_t* __restrict__ ncnt,  // [Tpad]
     float* __restrict__ out,           // [N] (path sum or score)
-    int64_t N, int32_t d, int32_t dpad, int32_t T, int32_t max_nodes,
-    float fT, float c_norm, int32_t finalize) {
+    int64_t N, int32_t d, int32_t dpad, int32_t Tpad, int32_t max_nodes,
+    int32_t height_limit, float fT, float c_norm, int32_t finalize) {
   const int tid = threadIdx.x;
+  const int rows_per_iter = RPT * 256;
 
-  int2* tlds = (int2*)smem;                // [ILP * max_nodes]
-  XT* rows = (XT*)(tlds + ILP * max_nodes);  // [256][dpad] if ROWS_LDS
+  int2* tlds = (int2*)smem;                    // [V4_ILP * max_nodes]
+  KT* rows = (KT*)(tlds + V4_ILP * max_nodes);  // [rows_per_iter][dpad]
 
-  for (int64_t block_row0 = (int64_t)blockIdx.x * 256; block_row0 < N;
-       block_row0 += (int64_t)gridDim.x * 256) {
-    const int64_t my_row = block_row0 + tid;
-    const int rows_here = (int)min((int64_t)256, N - block_row0);
+  for (int64_t block_row0 = (int64_t)blockIdx.x * rows_per_iter; block_row0 < N;
+       block_row0 += (int64_t)gridDim.x * rows_per_iter) {
+    const int rows_here = (int)min((int64_t)rows_per_iter, N - block_row0);
 
     if (ROWS_LDS) {
       __syncthreads();  // previous iteration's readers done
       const int64_t total = (int64_t)rows_here * d;
       for (int64_t g = tid; g < total; g += 256) {
-        const int r = (int)(g / d), c = (int)(g % d);
-        rows[r * dpad + c] = X[(block_row0 + r) * d + c];
+        const int r = (int)(g / (uint32_t)d), c = (int)(g % (uint32_t)d);
+        rows[r * dpad + c] = key_of_bits<KT>(X[(block_row0 + r) * d + c]);
       }
+      for (int r = tid; r < rows_per_iter; r += 256)
+        rows[r * dpad + d] = (KT)~(KT)0;  // leaf sentinel column
       __syncthreads();
     }
-    const XT* my_lrow = rows + tid * dpad;
 
-    float path_sum = 0.f;
-    for (int t = 0; t < T; t += ILP) {
-      __syncthreads();
-      const int live = min(ILP, T - t);
+    const KT* rb[RPT];
 #pragma unroll
-      for (int s = 0; s < ILP; ++s) {
-        if (s < live) {
-          const int ncs = ncnt[t + s];
-          const int2* ss = nodes + (int64_t)(t + s) * max_nodes;
-          for (int i = tid; i < ncs; i += 256) tlds[s * max_nodes + i] = ss[i];
-        }
+    for (int r = 0; r < RPT; ++r) rb[r] = rows + (tid + r * 256) * dpad;
+
+    float psum[RPT];
+#pragma unroll
+    for (int r = 0; r < RPT; ++r) psum[r] = 0.f;
+
+    for (int t = 0; t < Tpad; t += V4_ILP) {
+      __syncthreads();
+#pragma unroll
+      for (int s = 0; s < V4_ILP; ++s) {
+        const int ncs = ncnt[t + s];
+        const int2* ss = nodes + (int64_t)(t + s) * max_nodes;
+        for (int i = tid; i < ncs; i += 256) tlds[s * max_nodes + i] = ss[i];
       }
       __syncthreads();
 
-      if (my_row < N) {
-        int cur[ILP], dep[ILP];
-        float lf[ILP];
-        bool act[ILP];
-        bool any = false;
+      int cur[RPT][V4_ILP];
 #pragma unroll
-        for (int s = 0; s < ILP; ++s) {
-          cur[s] = 0;
-          dep[s] = 0;
-          lf[s] = 0.f;
-          act[s] = s < live;
-          any |= act[s];
-        }
-        while (any) {
-          int2 nd[ILP];
+      for (int r = 0; r < RPT; ++r)
 #pragma unroll
-          for (int s = 0; s < ILP; ++s) nd[s] = tlds[s * max_nodes + cur[s]];
-          any = false;
+        for (int s = 0; s < V4_ILP; ++s) cur[r][s] = 0;
+
+#pragma unroll 2
+      for (int it = 0; it < height_limit; ++it) {
+        int2 nd[RPT][V4_ILP];
 #pragma unroll
-          for (int s = 0; s < ILP; ++s) {
-            if (act[s]) {
-              if (nd[s].x < 0) {
-                lf[s] = __int_as_float(nd[s].y);
-                act[s] = false;
-              } else {
-                const float xv =
-                    ROWS_LDS
-                        ? cvt_feat(my_lrow[pn_feat(nd[s].x)])
-                        : load_feat<XT>(X, my_row * d + pn_feat(nd[s].x));
-                cur[s] =
-                    (xv < __int_as_float(nd[s].y)) ? cur[s] + 1 : pn_right(nd[s].x);
-                ++dep[s];
-                any = true;
-              }
+        for (int r = 0; r < RPT; ++r)
+#pragma unroll
+          for (int s = 0; s < V4_ILP; ++s)
+            nd[r][s] = tlds[s * max_nodes + cur[r][s]];
+#pragma unroll
+        for (int r = 0; r < RPT; ++r) {
+#pragma unroll
+          for (int s = 0; s < V4_ILP; ++s) {
+            const int f = pn_feat(nd[r][s].x);
+            uint32_t x;
+            if (ROWS_LDS) {
+              x = widen_key(rb[r][f]);
+            } else {
+              // global fallback: transform on the fly; the sentinel column
+              // does not exist in X, so guard the leaf feature index.
+              const int64_t my_row = block_row0 + tid + r * 256;
+              KT raw = (f < d && my_row < N) ? X[my_row * d + f] : (KT)0;
+              x = (f < d) ? widen_key(key_of_bits<KT>(raw)) : 0xFFFFFFFFu;
             }
+            cur[r][s] = (x < (uint32_t)nd[r][s].y) ? cur[r][s] + 1
+                                                   : pn_right(nd[r][s].x);
           }
         }
-        // oracle order (cpu_engine.path_lengths): per tree, in tree order:
-        // total = f32(total + f32(depth + leaf))
-#pragma unroll
-        for (int s = 0; s < ILP; ++s)
-          if (s < live)
-            path_sum = __fadd_rn(path_sum, __fadd_rn((float)dep[s], lf[s]));
       }
+      // every chain rests at its leaf (self-loop); read value = depth + c(m)
+#pragma unroll
+      for (int r = 0; r < RPT; ++r)
+#pragma unroll
+        for (int s = 0; s < V4_ILP; ++s)
+          psum[r] = __fadd_rn(
+              psum[r],
+              __int_as_float(tlds[s * max_nodes + cur[r][s]].y));
     }
-    if (my_row < N) {
-      if (finalize) {
-        const float mean32 = (float)((double)path_sum / (double)fT);
-        const double ratio = (double)mean32 / (double)c_norm;
-        out[my_row] = (float)exp2(-ratio);
-      } else {
-        out[my_row] = path_sum;
+
+#pragma unroll
+    for (int r = 0; r < RPT; ++r) {
+      const int64_t my_row = block_row0 + tid + r * 256;
+      if (my_row < N) {
+        if (finalize) {
+          const float mean32 = (float)((double)psum[r] / (double)fT);
+          const double ratio = (double)mean32 / (double)c_norm;
+          out[my_row] = (float)exp2(-ratio);
+        } else {
+          out[my_row] = psum[r];
+        }
       }
     }
   }
@@ -836,24 +886,28 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
                      nnz, max_nodes, height_limit);
 }
 
-void launch_score_forest(bool bf16, bool rows_lds, const void* X,
+void launch_score_forest(bool bf16, int rpt, bool rows_lds, const void* X,
                          const void* nodes, const int32_t* ncount, float* out,
-                         int64_t N, int32_t d, int32_t dpad, int32_t T,
-                         int32_t max_nodes, float fT, float c_norm,
-                         int finalize, size_t lds, int blocks,
+                         int64_t N, int32_t d, int32_t dpad, int32_t Tpad,
+                         int32_t max_nodes, int32_t height_limit, float fT,
+                         float c_norm, int finalize, size_t lds, int blocks,
                          hipStream_t stream) {
-#define LS(XT, RL)                                                            \
+#define LS(KT, RPT, RL)                                                       \
   do {                                                                        \
-    raise_lds((const void*)score_forest_kernel_t<XT, RL, 4>, lds);            \
-    hipLaunchKernelGGL((score_forest_kernel_t<XT, RL, 4>), dim3(blocks),      \
-                       dim3(256), lds, stream, (const XT*)X,                  \
-                       (const int2*)nodes, ncount, out, N, d, dpad, T,        \
-                       max_nodes, fT, c_norm, finalize);                      \
+    raise_lds((const void*)score_forest_v4<KT, RPT, RL>, lds);                \
+    hipLaunchKernelGGL((score_forest_v4<KT, RPT, RL>), dim3(blocks),          \
+                       dim3(256), lds, stream, (const KT*)X,                  \
+                       (const int2*)nodes, ncount, out, N, d, dpad, Tpad,     \
+                       max_nodes, height_limit, fT, c_norm, finalize);        \
   } while (0)
   if (bf16) {
-    if (rows_lds) LS(uint16_t, true); else LS(uint16_t, false);
+    if (rows_lds && rpt == 2) LS(uint16_t, 2, true);
+    else if (rows_lds) LS(uint16_t, 1, true);
+    else LS(uint16_t, 1, false);
   } else {
-    if (rows_lds) LS(float, true); else LS(float, false);
+    if (rows_lds && rpt == 2) LS(uint32_t, 2, true);
+    else if (rows_lds) LS(uint32_t, 1, true);
+    else LS(uint32_t, 1, false);
   }
 #undef LS
 }
