@@ -563,26 +563,42 @@ __global__ void __launch_bounds__(256) score_forest_v4(
 
 #pragma unroll 2
       for (int it = 0; it < height_limit; ++it) {
+        // three explicit phases so the compiler BATCHES the LDS reads
+        // (8 node reads -> one wait -> 8 key reads -> one wait -> VALU)
+        // instead of serializing 8 dependent read+wait round-trips.
         int2 nd[RPT][V4_ILP];
 #pragma unroll
         for (int r = 0; r < RPT; ++r)
 #pragma unroll
           for (int s = 0; s < V4_ILP; ++s)
             nd[r][s] = tlds[s * max_nodes + cur[r][s]];
+        KT kraw[RPT][V4_ILP];
 #pragma unroll
         for (int r = 0; r < RPT; ++r) {
 #pragma unroll
           for (int s = 0; s < V4_ILP; ++s) {
             const int f = pn_feat(nd[r][s].x);
+            if (ROWS_LDS) {
+              kraw[r][s] = rb[r][f];
+            } else {
+              // global fallback: the sentinel column does not exist in X,
+              // so guard the leaf feature index.
+              const int64_t my_row = block_row0 + tid + r * 256;
+              kraw[r][s] = (f < d && my_row < N) ? X[my_row * d + f] : (KT)0;
+            }
+          }
+        }
+#pragma unroll
+        for (int r = 0; r < RPT; ++r) {
+#pragma unroll
+          for (int s = 0; s < V4_ILP; ++s) {
             uint32_t x;
             if (ROWS_LDS) {
-              x = widen_key(rb[r][f]);
+              x = widen_key(kraw[r][s]);
             } else {
-              // global fallback: transform on the fly; the sentinel column
-              // does not exist in X, so guard the leaf feature index.
-              const int64_t my_row = block_row0 + tid + r * 256;
-              KT raw = (f < d && my_row < N) ? X[my_row * d + f] : (KT)0;
-              x = (f < d) ? widen_key(key_of_bits<KT>(raw)) : 0xFFFFFFFFu;
+              const int f = pn_feat(nd[r][s].x);
+              x = (f < d) ? widen_key(key_of_bits<KT>(kraw[r][s]))
+                          : 0xFFFFFFFFu;
             }
             cur[r][s] = (x < (uint32_t)nd[r][s].y) ? cur[r][s] + 1
                                                    : pn_right(nd[r][s].x);
