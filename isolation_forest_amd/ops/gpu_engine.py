@@ -225,6 +225,38 @@ def _nodes_packed_v4(forest, d_sentinel: int, bf16: bool):
     return packed, ncount, max(max_depth, 1)
 
 
+def _eif_dense_packed(forest):
+    """Packing for score_extended_dense_v2: w0 = right<<12 (leaf/pad: own id
+    -> self-loop), w1 = offset f32 (leaf/pad: -inf so the zero weight row's
+    dot=0 routes right), plus a depth-folded leaf-value array.
+    Returns (packed int32 [T, mn, 2], values f32 [T, mn], max_depth)."""
+    T, mn = forest.feature.shape
+    if mn > 32767:
+        raise ValueError("forest too deep for the packed node format")
+    feat = forest.feature
+    internal = feat >= 0
+    leaf = feat == ExtendedForest.LEAF
+    depth = _node_depths(feat, forest.right)
+    ids = np.broadcast_to(np.arange(mn, dtype=np.int32)[None, :], (T, mn))
+    w0 = np.where(internal, forest.right.astype(np.int32), ids) << 12
+    w1 = np.where(
+        internal,
+        forest.value.astype(np.float32),
+        np.float32(-np.inf),
+    ).astype(np.float32)
+    packed = np.empty((T, mn, 2), dtype=np.int32)
+    packed[..., 0] = w0
+    packed[..., 1] = w1.view(np.int32)
+    values = np.where(
+        leaf,
+        depth.astype(np.float32) + forest.value.astype(np.float32),
+        np.float32(0.0),
+    ).astype(np.float32)
+    live = internal | leaf
+    max_depth = int(depth[live].max()) if live.any() else 0
+    return packed, values, max(max_depth, 1)
+
+
 def _nodes_packed(forest) -> np.ndarray:
     """Packed 8-byte node records for the EIF scoring kernels:
     meta<0 => leaf (value = c(count)); else feature = meta&0xFFF,
@@ -255,7 +287,18 @@ def _device_forest(model, device, v4_key=None):
     if key not in cache:
         forest = model.forest
         extra = {}
-        if v4_key is not None:
+        if v4_key == "eif_dense":
+            packed, values, max_depth = _eif_dense_packed(forest)
+            aos = torch.from_numpy(packed).to(device)
+            ncount = torch.from_numpy(
+                np.ascontiguousarray(forest.node_count, dtype=np.int32)
+            ).to(device)
+            extra["values"] = torch.from_numpy(values).to(device)
+            extra["hw"] = torch.from_numpy(
+                np.ascontiguousarray(forest.hyper_w)
+            ).to(device)
+            extra["height"] = max_depth
+        elif v4_key is not None:
             d, bf16 = v4_key
             packed, ncount_np, max_depth = _nodes_packed_v4(forest, d, bf16)
             aos = torch.from_numpy(packed).to(device)
@@ -294,8 +337,15 @@ def score_forest(model, X: torch.Tensor, finalize: bool = True) -> torch.Tensor:
 def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torch.Tensor:
     ext = load_extension()
     forest = model.forest
-    aos, ncount, extra = _device_forest(model, X.device)
     c = float(avg_path_length(forest.num_samples))
+    d = int(X.shape[1])
+    if forest.nnz == d and d <= 32:
+        aos, ncount, extra = _device_forest(model, X.device, v4_key="eif_dense")
+        return ext.score_extended_dense_v2(
+            X.contiguous(), aos, extra["values"], extra["hw"], ncount,
+            extra["height"], c, finalize,
+        )
+    aos, ncount, extra = _device_forest(model, X.device)
     return ext.score_extended_forest(
         X.contiguous(), aos, extra["hidx"], extra["hw"], ncount, c, finalize
     )

@@ -56,6 +56,15 @@ void launch_score_extended_forest(bool bf16, bool rows_lds, bool hyper_lds,
                                   int finalize, size_t lds, int blocks,
                                   hipStream_t stream);
 
+void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
+                                    const void* nodes, const float* values,
+                                    const float* hw, const int32_t* ncount,
+                                    float* out, int64_t N, int32_t d,
+                                    int32_t T, int32_t max_nodes,
+                                    int32_t height_limit, float fT,
+                                    float c_norm, int finalize, size_t lds,
+                                    int blocks, hipStream_t stream);
+
 }  // namespace ifa
 
 namespace {
@@ -305,6 +314,47 @@ torch::Tensor score_extended_forest(torch::Tensor X,
   return out;
 }
 
+torch::Tensor score_extended_dense_v2(torch::Tensor X,
+                                      torch::Tensor nodes_packed,
+                                      torch::Tensor values, torch::Tensor hw,
+                                      torch::Tensor ncount,
+                                      int64_t height_limit, double c_norm,
+                                      bool finalize) {
+  CHECK_CUDA(X);
+  CHECK_CONTIG(X);
+  CHECK_CUDA(nodes_packed);
+  CHECK_CONTIG(nodes_packed);
+  CHECK_CUDA(values);
+  CHECK_CONTIG(values);
+  CHECK_CUDA(hw);
+  CHECK_CONTIG(hw);
+  CHECK_CUDA(ncount);
+  check_x(X);
+  TORCH_CHECK(nodes_packed.dim() == 3 && nodes_packed.size(2) == 2 &&
+                  nodes_packed.scalar_type() == torch::kInt32,
+              "nodes must be packed int32 [T, max_nodes, 2]");
+  TORCH_CHECK(hw.dim() == 3 && hw.scalar_type() == torch::kFloat32,
+              "hw must be float32 [T, max_nodes, d]");
+  int64_t N = X.size(0), d = X.size(1);
+  int64_t T = nodes_packed.size(0), max_nodes = nodes_packed.size(1);
+  TORCH_CHECK(hw.size(2) == d, "dense v2 requires nnz == d");
+  TORCH_CHECK(d <= 32, "dense v2 supports d <= 32");
+  auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
+  if (N == 0) return out;
+
+  const int D = d <= 8 ? 8 : (d <= 16 ? 16 : 32);
+  size_t lds = (size_t)max_nodes * (8 + 4 + (D + 4) * 4);
+  TORCH_CHECK(lds <= kMaxLds, "tree too large for LDS staging");
+  int blocks = (int)std::min<int64_t>((N + 511) / 512, 8192);
+  ifa::launch_score_extended_dense_v2(
+      is_bf16(X), D, X.data_ptr(), nodes_packed.data_ptr<int32_t>(),
+      values.data_ptr<float>(), hw.data_ptr<float>(),
+      ncount.data_ptr<int32_t>(), out.data_ptr<float>(), N, (int32_t)d,
+      (int32_t)T, (int32_t)max_nodes, (int32_t)height_limit, (float)T,
+      (float)c_norm, finalize ? 1 : 0, lds, blocks, current_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bag_gather", &bag_gather, "gather per-tree bags (K9/K10)");
   m.def("build_forest", &build_forest, "build standard iTrees (K1/K2/K11)");
@@ -313,5 +363,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("score_forest", &score_forest, "batched path-length scoring (K6)");
   m.def("score_extended_forest", &score_extended_forest,
         "batched EIF scoring (K7)");
+  m.def("score_extended_dense_v2", &score_extended_dense_v2,
+        "dense EIF scoring, rows-in-registers (K7 fast path)");
   m.attr("WAVE") = 64;
 }

@@ -751,6 +751,146 @@ __global__ void __launch_bounds__(256) score_extended_dense_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// extended scoring, dense fast path v2 (nnz == d, d <= 32).
+//
+// Rows live in REGISTERS (f32, templated D) across the whole tree loop —
+// loaded once per 512-row block iteration, reused for all T trees. Per tree
+// the node records {right|self-loop, offset32 / -inf}, the depth-folded leaf
+// values and the dense weight matrix (row stride D+4 words to spread the
+// b128 reads across banks) are staged in LDS. Each thread walks RPT rows
+// concurrently through a FIXED height-trip loop (leaves self-loop: zero
+// weight row + offset -inf routes right). The dot uses 4 partial fma
+// accumulators (same reassociation contract as v1 — tolerance, not
+// bitwise; build stays bitwise).
+// ---------------------------------------------------------------------------
+
+template <typename KT>  // u16 = bf16 rows, u32 = f32 rows
+__device__ __forceinline__ float load_row_f32(const KT* X, int64_t off);
+template <>
+__device__ __forceinline__ float load_row_f32<uint16_t>(const uint16_t* X,
+                                                        int64_t off) {
+  return cvt_feat(X[off]);
+}
+template <>
+__device__ __forceinline__ float load_row_f32<uint32_t>(const uint32_t* X,
+                                                        int64_t off) {
+  union { uint32_t u; float f; } cv;
+  cv.u = X[off];
+  return cv.f;
+}
+
+template <typename KT, int D, int RPT>
+__global__ void __launch_bounds__(256) score_extended_dense_v2(
+    const KT* __restrict__ X,           // raw bits [N][d]
+    const int2* __restrict__ nodes,     // [T][max_nodes] {w0, offset/-inf}
+    const float* __restrict__ values,   // [T][max_nodes] leaf value+depth
+    const float* __restrict__ hw,       // [T][max_nodes][d] dense weights
+    const int32_t* __restrict__ ncnt,   // [T]
+    float* __restrict__ out, int64_t N, int32_t d, int32_t T,
+    int32_t max_nodes, int32_t height_limit, float fT, float c_norm,
+    int32_t finalize) {
+  const int tid = threadIdx.x;
+  const int rows_per_iter = RPT * 256;
+  const int DW = D + 4;  // LDS weight-row stride in words
+
+  int2* tlds = (int2*)smem;                  // [max_nodes]
+  float* vlds = (float*)(tlds + max_nodes);  // [max_nodes]
+  float* wlds = vlds + max_nodes;            // [max_nodes][DW]
+
+  for (int64_t block_row0 = (int64_t)blockIdx.x * rows_per_iter; block_row0 < N;
+       block_row0 += (int64_t)gridDim.x * rows_per_iter) {
+    float row[RPT][D];
+#pragma unroll
+    for (int r = 0; r < RPT; ++r) {
+      const int64_t my_row = block_row0 + tid + r * 256;
+      const bool ok = my_row < N;
+#pragma unroll
+      for (int j = 0; j < D; ++j)
+        row[r][j] = (ok && j < d) ? load_row_f32<KT>(X, my_row * d + j) : 0.f;
+    }
+
+    float psum[RPT];
+#pragma unroll
+    for (int r = 0; r < RPT; ++r) psum[r] = 0.f;
+
+    for (int t = 0; t < T; ++t) {
+      __syncthreads();
+      const int nc = ncnt[t];
+      {
+        const int2* ss = nodes + (int64_t)t * max_nodes;
+        const float* vs = values + (int64_t)t * max_nodes;
+        for (int i = tid; i < nc; i += 256) {
+          tlds[i] = ss[i];
+          vlds[i] = vs[i];
+        }
+        const float* ws = hw + (int64_t)t * max_nodes * d;
+        const int total = nc * d;
+        for (int g = tid; g < total; g += 256) {
+          const int i = g / (uint32_t)d, j = g % (uint32_t)d;
+          wlds[i * DW + j] = ws[g];
+        }
+        if (d < D) {
+          const int padn = nc * (D - d);
+          for (int g = tid; g < padn; g += 256) {
+            const int i = g / (uint32_t)(D - d), j = g % (uint32_t)(D - d);
+            wlds[i * DW + d + j] = 0.f;
+          }
+        }
+      }
+      __syncthreads();
+
+      int cur[RPT];
+#pragma unroll
+      for (int r = 0; r < RPT; ++r) cur[r] = 0;
+
+      for (int it = 0; it < height_limit; ++it) {
+        int2 nd[RPT];
+#pragma unroll
+        for (int r = 0; r < RPT; ++r) nd[r] = tlds[cur[r]];
+        float4 w[RPT][D / 4];
+#pragma unroll
+        for (int r = 0; r < RPT; ++r) {
+          const float4* wp = (const float4*)(wlds + cur[r] * DW);
+#pragma unroll
+          for (int j4 = 0; j4 < D / 4; ++j4) w[r][j4] = wp[j4];
+        }
+#pragma unroll
+        for (int r = 0; r < RPT; ++r) {
+          float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+#pragma unroll
+          for (int j4 = 0; j4 < D / 4; ++j4) {
+            a0 = __builtin_fmaf(w[r][j4].x, row[r][4 * j4 + 0], a0);
+            a1 = __builtin_fmaf(w[r][j4].y, row[r][4 * j4 + 1], a1);
+            a2 = __builtin_fmaf(w[r][j4].z, row[r][4 * j4 + 2], a2);
+            a3 = __builtin_fmaf(w[r][j4].w, row[r][4 * j4 + 3], a3);
+          }
+          const float dot = __fadd_rn(__fadd_rn(a0, a1), __fadd_rn(a2, a3));
+          cur[r] = (dot < __int_as_float(nd[r].y)) ? cur[r] + 1
+                                                   : pn_right(nd[r].x);
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < RPT; ++r)
+        psum[r] = __fadd_rn(psum[r], vlds[cur[r]]);
+    }
+
+#pragma unroll
+    for (int r = 0; r < RPT; ++r) {
+      const int64_t my_row = block_row0 + tid + r * 256;
+      if (my_row < N) {
+        if (finalize) {
+          const float mean32 = (float)((double)psum[r] / (double)fT);
+          const double ratio = (double)mean32 / (double)c_norm;
+          out[my_row] = (float)exp2(-ratio);
+        } else {
+          out[my_row] = psum[r];
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // extended scoring, sparse general path (nnz < d): exact oracle order.
 // ---------------------------------------------------------------------------
 
@@ -955,6 +1095,34 @@ void launch_score_extended_dense(bool bf16, bool rows_lds, bool wlds,
     else LSD(float, false, false);
   }
 #undef LSD
+}
+
+void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
+                                    const void* nodes, const float* values,
+                                    const float* hw, const int32_t* ncount,
+                                    float* out, int64_t N, int32_t d,
+                                    int32_t T, int32_t max_nodes,
+                                    int32_t height_limit, float fT,
+                                    float c_norm, int finalize, size_t lds,
+                                    int blocks, hipStream_t stream) {
+#define LSD2(KT, DD)                                                          \
+  do {                                                                        \
+    raise_lds((const void*)score_extended_dense_v2<KT, DD, 2>, lds);          \
+    hipLaunchKernelGGL((score_extended_dense_v2<KT, DD, 2>), dim3(blocks),    \
+                       dim3(256), lds, stream, (const KT*)X,                  \
+                       (const int2*)nodes, values, hw, ncount, out, N, d, T,  \
+                       max_nodes, height_limit, fT, c_norm, finalize);        \
+  } while (0)
+  if (bf16) {
+    if (D == 8) LSD2(uint16_t, 8);
+    else if (D == 16) LSD2(uint16_t, 16);
+    else LSD2(uint16_t, 32);
+  } else {
+    if (D == 8) LSD2(uint32_t, 8);
+    else if (D == 16) LSD2(uint32_t, 16);
+    else LSD2(uint32_t, 32);
+  }
+#undef LSD2
 }
 
 void launch_score_extended_forest(bool bf16, bool rows_lds, bool hyper_lds,
