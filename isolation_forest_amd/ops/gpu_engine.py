@@ -294,9 +294,16 @@ def _device_forest(model, device, v4_key=None):
                 np.ascontiguousarray(forest.node_count, dtype=np.int32)
             ).to(device)
             extra["values"] = torch.from_numpy(values).to(device)
-            extra["hw"] = torch.from_numpy(
-                np.ascontiguousarray(forest.hyper_w)
-            ).to(device)
+            # pad weights to the kernel's compile-time D columns
+            d = forest.hyper_w.shape[2]
+            D = 8 if d <= 8 else (16 if d <= 16 else 32)
+            hw = forest.hyper_w
+            if d < D:
+                T, mn, _ = hw.shape
+                hw = np.concatenate(
+                    [hw, np.zeros((T, mn, D - d), dtype=np.float32)], axis=2
+                )
+            extra["hw"] = torch.from_numpy(np.ascontiguousarray(hw)).to(device)
             extra["height"] = max_depth
         elif v4_key is not None:
             d, bf16 = v4_key

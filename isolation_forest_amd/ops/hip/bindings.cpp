@@ -337,13 +337,13 @@ torch::Tensor score_extended_dense_v2(torch::Tensor X,
               "hw must be float32 [T, max_nodes, d]");
   int64_t N = X.size(0), d = X.size(1);
   int64_t T = nodes_packed.size(0), max_nodes = nodes_packed.size(1);
-  TORCH_CHECK(hw.size(2) == d, "dense v2 requires nnz == d");
   TORCH_CHECK(d <= 32, "dense v2 supports d <= 32");
+  const int D = d <= 8 ? 8 : (d <= 16 ? 16 : 32);
+  TORCH_CHECK(hw.size(2) == D, "hw must be host-padded to D columns");
   auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
   if (N == 0) return out;
 
-  const int D = d <= 8 ? 8 : (d <= 16 ? 16 : 32);
-  size_t lds = (size_t)max_nodes * (8 + 4 + (D + 4) * 4);
+  size_t lds = (size_t)max_nodes * 12 + 16 + (size_t)max_nodes * (D / 4 + 1) * 16;
   TORCH_CHECK(lds <= kMaxLds, "tree too large for LDS staging");
   int blocks = (int)std::min<int64_t>((N + 511) / 512, 8192);
   ifa::launch_score_extended_dense_v2(

@@ -779,34 +779,75 @@ __device__ __forceinline__ float load_row_f32<uint32_t>(const uint32_t* X,
   return cv.f;
 }
 
+// Row storage for the dense-EIF walk: bf16 rows stay PACKED (2 per VGPR,
+// bf16->f32 unpack is an exact shift/mask), f32 rows are stored directly.
+template <typename KT, int D>
+struct RowReg;
+
+template <int D>
+struct RowReg<uint16_t, D> {
+  uint32_t pk[D / 2];
+  __device__ __forceinline__ void load(const uint16_t* X, int64_t base,
+                                       int d, bool ok) {
+#pragma unroll
+    for (int p = 0; p < D / 2; ++p) {
+      const uint32_t lo =
+          (ok && 2 * p < d) ? (uint32_t)X[base + 2 * p] : 0u;
+      const uint32_t hi =
+          (ok && 2 * p + 1 < d) ? (uint32_t)X[base + 2 * p + 1] : 0u;
+      pk[p] = lo | (hi << 16);
+    }
+  }
+  __device__ __forceinline__ float get(int j) const {  // j compile-time
+    union { uint32_t u; float f; } cv;
+    cv.u = (j & 1) ? (pk[j / 2] & 0xFFFF0000u) : (pk[j / 2] << 16);
+    return cv.f;
+  }
+};
+
+template <int D>
+struct RowReg<uint32_t, D> {
+  uint32_t pk[D];
+  __device__ __forceinline__ void load(const uint32_t* X, int64_t base,
+                                       int d, bool ok) {
+#pragma unroll
+    for (int j = 0; j < D; ++j)
+      pk[j] = (ok && j < d) ? X[base + j] : 0u;
+  }
+  __device__ __forceinline__ float get(int j) const {
+    union { uint32_t u; float f; } cv;
+    cv.u = pk[j];
+    return cv.f;
+  }
+};
+
+#define EIFD_THREADS 512
+
 template <typename KT, int D, int RPT>
-__global__ void __launch_bounds__(256) score_extended_dense_v2(
+__global__ void __launch_bounds__(EIFD_THREADS, 4) score_extended_dense_v2(
     const KT* __restrict__ X,           // raw bits [N][d]
     const int2* __restrict__ nodes,     // [T][max_nodes] {w0, offset/-inf}
     const float* __restrict__ values,   // [T][max_nodes] leaf value+depth
-    const float* __restrict__ hw,       // [T][max_nodes][d] dense weights
+    const float* __restrict__ hw,       // [T][max_nodes][D] PADDED dense w
     const int32_t* __restrict__ ncnt,   // [T]
     float* __restrict__ out, int64_t N, int32_t d, int32_t T,
     int32_t max_nodes, int32_t height_limit, float fT, float c_norm,
     int32_t finalize) {
   const int tid = threadIdx.x;
-  const int rows_per_iter = RPT * 256;
-  const int DW = D + 4;  // LDS weight-row stride in words
+  const int rows_per_iter = RPT * EIFD_THREADS;
+  const int DW4 = D / 4 + 1;  // LDS weight-row stride in float4s
 
   int2* tlds = (int2*)smem;                  // [max_nodes]
   float* vlds = (float*)(tlds + max_nodes);  // [max_nodes]
-  float* wlds = vlds + max_nodes;            // [max_nodes][DW]
+  float4* wlds = (float4*)(((uintptr_t)(vlds + max_nodes) + 15) & ~15ull);
 
   for (int64_t block_row0 = (int64_t)blockIdx.x * rows_per_iter; block_row0 < N;
        block_row0 += (int64_t)gridDim.x * rows_per_iter) {
-    float row[RPT][D];
+    RowReg<KT, D> row[RPT];
 #pragma unroll
     for (int r = 0; r < RPT; ++r) {
-      const int64_t my_row = block_row0 + tid + r * 256;
-      const bool ok = my_row < N;
-#pragma unroll
-      for (int j = 0; j < D; ++j)
-        row[r][j] = (ok && j < d) ? load_row_f32<KT>(X, my_row * d + j) : 0.f;
+      const int64_t my_row = block_row0 + tid + r * EIFD_THREADS;
+      row[r].load(X, my_row * d, d, my_row < N);
     }
 
     float psum[RPT];
@@ -819,22 +860,17 @@ __global__ void __launch_bounds__(256) score_extended_dense_v2(
       {
         const int2* ss = nodes + (int64_t)t * max_nodes;
         const float* vs = values + (int64_t)t * max_nodes;
-        for (int i = tid; i < nc; i += 256) {
+        for (int i = tid; i < nc; i += EIFD_THREADS) {
           tlds[i] = ss[i];
           vlds[i] = vs[i];
         }
-        const float* ws = hw + (int64_t)t * max_nodes * d;
-        const int total = nc * d;
-        for (int g = tid; g < total; g += 256) {
-          const int i = g / (uint32_t)d, j = g % (uint32_t)d;
-          wlds[i * DW + j] = ws[g];
-        }
-        if (d < D) {
-          const int padn = nc * (D - d);
-          for (int g = tid; g < padn; g += 256) {
-            const int i = g / (uint32_t)(D - d), j = g % (uint32_t)(D - d);
-            wlds[i * DW + d + j] = 0.f;
-          }
+        // weights are host-padded to D columns: pure float4 copy with
+        // shift indexing (D/4 is a power of two for D in {8,16,32})
+        const float4* ws = (const float4*)(hw + (int64_t)t * max_nodes * D);
+        const int total4 = nc * (D / 4);
+        for (int g = tid; g < total4; g += EIFD_THREADS) {
+          const int i = g / (D / 4), j4 = g % (D / 4);
+          wlds[i * DW4 + j4] = ws[g];
         }
       }
       __syncthreads();
@@ -844,29 +880,46 @@ __global__ void __launch_bounds__(256) score_extended_dense_v2(
       for (int r = 0; r < RPT; ++r) cur[r] = 0;
 
       for (int it = 0; it < height_limit; ++it) {
+        // phase 1: batched node reads; derive both successors up front so
+        // the select stays branchless (no exec-mask divergence)
         int2 nd[RPT];
 #pragma unroll
         for (int r = 0; r < RPT; ++r) nd[r] = tlds[cur[r]];
-        float4 w[RPT][D / 4];
+        // phases 2+3: weight reads in chunks to bound live registers
+        // (quarter chunks at D=32 keep the kernel under 128 VGPRs)
+        constexpr int CHUNKS = (D >= 32) ? 4 : 2;
+        constexpr int CW4 = D / (4 * CHUNKS);  // float4s per chunk
+        float a0[RPT], a1[RPT], a2[RPT], a3[RPT];
 #pragma unroll
-        for (int r = 0; r < RPT; ++r) {
-          const float4* wp = (const float4*)(wlds + cur[r] * DW);
+        for (int r = 0; r < RPT; ++r) a0[r] = a1[r] = a2[r] = a3[r] = 0.f;
 #pragma unroll
-          for (int j4 = 0; j4 < D / 4; ++j4) w[r][j4] = wp[j4];
+        for (int ch = 0; ch < CHUNKS; ++ch) {
+          float4 w[RPT][CW4];
+#pragma unroll
+          for (int r = 0; r < RPT; ++r) {
+            const float4* wp = wlds + cur[r] * DW4 + ch * CW4;
+#pragma unroll
+            for (int j4 = 0; j4 < CW4; ++j4) w[r][j4] = wp[j4];
+          }
+#pragma unroll
+          for (int r = 0; r < RPT; ++r) {
+#pragma unroll
+            for (int j4 = 0; j4 < CW4; ++j4) {
+              const int j = ch * (D / CHUNKS) + 4 * j4;
+              a0[r] = __builtin_fmaf(w[r][j4].x, row[r].get(j + 0), a0[r]);
+              a1[r] = __builtin_fmaf(w[r][j4].y, row[r].get(j + 1), a1[r]);
+              a2[r] = __builtin_fmaf(w[r][j4].z, row[r].get(j + 2), a2[r]);
+              a3[r] = __builtin_fmaf(w[r][j4].w, row[r].get(j + 3), a3[r]);
+            }
+          }
         }
 #pragma unroll
         for (int r = 0; r < RPT; ++r) {
-          float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
-#pragma unroll
-          for (int j4 = 0; j4 < D / 4; ++j4) {
-            a0 = __builtin_fmaf(w[r][j4].x, row[r][4 * j4 + 0], a0);
-            a1 = __builtin_fmaf(w[r][j4].y, row[r][4 * j4 + 1], a1);
-            a2 = __builtin_fmaf(w[r][j4].z, row[r][4 * j4 + 2], a2);
-            a3 = __builtin_fmaf(w[r][j4].w, row[r][4 * j4 + 3], a3);
-          }
-          const float dot = __fadd_rn(__fadd_rn(a0, a1), __fadd_rn(a2, a3));
-          cur[r] = (dot < __int_as_float(nd[r].y)) ? cur[r] + 1
-                                                   : pn_right(nd[r].x);
+          const float dot = __fadd_rn(__fadd_rn(a0[r], a1[r]),
+                                      __fadd_rn(a2[r], a3[r]));
+          const int right = pn_right(nd[r].x);
+          const int left = cur[r] + 1;
+          cur[r] = (dot < __int_as_float(nd[r].y)) ? left : right;
         }
       }
 #pragma unroll
@@ -876,7 +929,7 @@ __global__ void __launch_bounds__(256) score_extended_dense_v2(
 
 #pragma unroll
     for (int r = 0; r < RPT; ++r) {
-      const int64_t my_row = block_row0 + tid + r * 256;
+      const int64_t my_row = block_row0 + tid + r * EIFD_THREADS;
       if (my_row < N) {
         if (finalize) {
           const float mean32 = (float)((double)psum[r] / (double)fT);
@@ -1105,22 +1158,24 @@ void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
                                     int32_t height_limit, float fT,
                                     float c_norm, int finalize, size_t lds,
                                     int blocks, hipStream_t stream) {
-#define LSD2(KT, DD)                                                          \
+#define LSD2(KT, DD, RR)                                                      \
   do {                                                                        \
-    raise_lds((const void*)score_extended_dense_v2<KT, DD, 2>, lds);          \
-    hipLaunchKernelGGL((score_extended_dense_v2<KT, DD, 2>), dim3(blocks),    \
-                       dim3(256), lds, stream, (const KT*)X,                  \
+    raise_lds((const void*)score_extended_dense_v2<KT, DD, RR>, lds);         \
+    hipLaunchKernelGGL((score_extended_dense_v2<KT, DD, RR>), dim3(blocks),   \
+                       dim3(EIFD_THREADS), lds, stream, (const KT*)X,         \
                        (const int2*)nodes, values, hw, ncount, out, N, d, T,  \
                        max_nodes, height_limit, fT, c_norm, finalize);        \
   } while (0)
+  // D=32 runs RPT=1: 2 rows/thread would spill past the 128-VGPR budget
+  // that keeps 4 waves/SIMD resident (2 blocks/CU at ~80 KB LDS each)
   if (bf16) {
-    if (D == 8) LSD2(uint16_t, 8);
-    else if (D == 16) LSD2(uint16_t, 16);
-    else LSD2(uint16_t, 32);
+    if (D == 8) LSD2(uint16_t, 8, 2);
+    else if (D == 16) LSD2(uint16_t, 16, 2);
+    else LSD2(uint16_t, 32, 1);
   } else {
-    if (D == 8) LSD2(uint32_t, 8);
-    else if (D == 16) LSD2(uint32_t, 16);
-    else LSD2(uint32_t, 32);
+    if (D == 8) LSD2(uint32_t, 8, 2);
+    else if (D == 16) LSD2(uint32_t, 16, 2);
+    else LSD2(uint32_t, 32, 1);
   }
 #undef LSD2
 }
