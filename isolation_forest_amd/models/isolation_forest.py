@@ -18,6 +18,7 @@ import torch
 from ..core import cpu_engine
 from ..core.forest import Forest
 from ..utils.params import Params, resolve_params
+from ..utils.timing import PhaseTimes, phase
 from . import base
 from .base import ModelBase, new_uid
 
@@ -68,6 +69,7 @@ class IsolationForest:
         for a fixed seed (when each rank sees the same rows) and
         statistically identical under row sharding.
         """
+        times = PhaseTimes()
         X, kind = base.extract_features(data, self.params.get("featuresCol"))
         base.check_output_columns(
             data, kind, self.params.get("scoreCol"), self.params.get("predictionCol")
@@ -75,11 +77,12 @@ class IsolationForest:
         seed = self.params.get("randomSeed")
         n_local = X.shape[0]
         total_features = X.shape[1]
-        if comm is not None:
-            total_rows = comm.all_reduce_sum_int(n_local)
-        else:
-            total_rows = n_local
-        rp = resolve_params(self.params, total_rows, total_features)
+        with phase(times, "resolve"):
+            if comm is not None:
+                total_rows = comm.all_reduce_sum_int(n_local)
+            else:
+                total_rows = n_local
+            rp = resolve_params(self.params, total_rows, total_features)
         if comm is not None and rp.num_samples > n_local:
             raise ValueError(
                 f"resolved maxSamples {rp.num_samples} exceeds this rank's local "
@@ -96,16 +99,21 @@ class IsolationForest:
         world = comm.world_size if comm is not None else 1
         t_lo, t_hi = _shard_range(T, rank, world)
 
-        forest_local = self._build_local(X, rp, seed, t_lo, t_hi, n_local)
-        forest = self._gather_forest(forest_local, comm, rp, total_features)
+        with phase(times, "bag+build", device=X if isinstance(X, torch.Tensor) else None):
+            forest_local = self._build_local(X, rp, seed, t_lo, t_hi, n_local)
+        with phase(times, "gather"):
+            forest = self._gather_forest(forest_local, comm, rp, total_features)
 
         model = self._model_cls()(
             uid=self.uid, forest=forest, params=self.params.copy()
         )
         model._resolved = rp
         if self.params.get("contamination") > 0.0:
-            scores = model.score(X)
-            base.fit_threshold(model, scores, self.params, comm=comm)
+            with phase(times, "threshold", device=X if isinstance(X, torch.Tensor) else None):
+                scores = model.score(X)
+                base.fit_threshold(model, scores, self.params, comm=comm)
+        model.fit_metrics = times
+        times.log(f"fit[{self.uid}]")
         return model
 
     # -- internals -------------------------------------------------------
