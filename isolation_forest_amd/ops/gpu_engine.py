@@ -208,7 +208,7 @@ def _nodes_packed_v4(forest, d_sentinel: int, bf16: bool):
         w1[internal] = _bf16_threshold_keys(s) if bf16 else _key32(s)
     leafval = (depth.astype(np.float32) + val.astype(np.float32))
     w1[leaf] = leafval[leaf].astype(np.float32).view(np.uint32)
-    pad = (-T) % 4
+    pad = (-T) % 8
     if pad:
         w0p = np.empty((pad, mn), dtype=np.int32)
         w0p[:] = np.int32(d_sentinel) | (ids[0] << 12)

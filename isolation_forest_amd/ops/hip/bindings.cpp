@@ -32,9 +32,10 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
                                   int32_t max_nodes, int32_t height_limit,
                                   size_t lds, hipStream_t stream);
 
-void launch_score_forest(bool bf16, int rpt, bool rows_lds, const void* X,
-                         const void* nodes, const int32_t* ncount, float* out,
-                         int64_t N, int32_t d, int32_t dpad, int32_t Tpad,
+void launch_score_forest(bool bf16, int rpt, bool rows_lds, int ilp,
+                         const void* X, const void* nodes,
+                         const int32_t* ncount, float* out, int64_t N,
+                         int32_t d, int32_t dpad, int32_t Tpad,
                          int32_t max_nodes, int32_t height_limit, float fT,
                          float c_norm, int finalize, size_t lds, int blocks,
                          hipStream_t stream);
@@ -203,7 +204,11 @@ torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
               "nodes must be packed int32 [Tpad, max_nodes, 2]");
   int64_t N = X.size(0), d = X.size(1);
   int64_t Tpad = nodes_packed.size(0), max_nodes = nodes_packed.size(1);
-  TORCH_CHECK(Tpad % 4 == 0, "packed tree count must be padded to 4");
+  TORCH_CHECK(Tpad % 8 == 0, "packed tree count must be padded to 8");
+  // staged-tree ILP: 4 (default) or 8; 8 halves the number of barrier-
+  // bounded staging phases at the cost of occupancy (A/B via env)
+  int ilp = 4;
+  if (const char* e = getenv("IFA_SCORE_ILP")) ilp = atoi(e) == 8 ? 8 : 4;
   TORCH_CHECK(d <= 4094, "scoring supports d <= 4094 (12-bit feature field "
                          "plus the leaf sentinel column)");
   auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
@@ -219,7 +224,7 @@ torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
   } else {
     while (dpad % 2 != 1) ++dpad;  // odd word count
   }
-  const size_t node_bytes = (size_t)4 * max_nodes * 8;  // ILP=4 staged trees
+  const size_t node_bytes = (size_t)ilp * max_nodes * 8;  // staged trees
   TORCH_CHECK(node_bytes <= kMaxLds, "tree too large for LDS staging");
   // prefer 2 rows/thread when 3 blocks/CU still fit, else 1, else global X
   int rpt = 2;
@@ -237,7 +242,7 @@ torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
   int64_t rows_per_block = (int64_t)(rows_lds ? rpt : 1) * 256;
   int blocks = (int)std::min<int64_t>(
       (N + rows_per_block - 1) / rows_per_block, 8192);
-  ifa::launch_score_forest(bf16, rpt, rows_lds, X.data_ptr(),
+  ifa::launch_score_forest(bf16, rpt, rows_lds, ilp, X.data_ptr(),
                            nodes_packed.data_ptr<int32_t>(),
                            ncount.data_ptr<int32_t>(), out.data_ptr<float>(),
                            N, (int32_t)d, (int32_t)dpad, (int32_t)Tpad,

@@ -505,9 +505,8 @@ __device__ __host__ __forceinline__ int row_stride(int d) {
   return d + pad;
 }
 
-#define V4_ILP 4
 
-template <typename KT, int RPT, bool ROWS_LDS>
+template <typename KT, int RPT, bool ROWS_LDS, int V4_ILP>
 __global__ void __launch_bounds__(256) score_forest_v4(
     const KT* __restrict__ X,          // raw bf16/f32 bits [N][d]
     const int2* __restrict__ nodes,    // [Tpad][max_nodes] packed v4
@@ -1095,29 +1094,35 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
                      nnz, max_nodes, height_limit);
 }
 
-void launch_score_forest(bool bf16, int rpt, bool rows_lds, const void* X,
-                         const void* nodes, const int32_t* ncount, float* out,
-                         int64_t N, int32_t d, int32_t dpad, int32_t Tpad,
+void launch_score_forest(bool bf16, int rpt, bool rows_lds, int ilp,
+                         const void* X, const void* nodes,
+                         const int32_t* ncount, float* out, int64_t N,
+                         int32_t d, int32_t dpad, int32_t Tpad,
                          int32_t max_nodes, int32_t height_limit, float fT,
                          float c_norm, int finalize, size_t lds, int blocks,
                          hipStream_t stream) {
-#define LS(KT, RPT, RL)                                                       \
+#define LS(KT, RPT, RL, TI)                                                   \
   do {                                                                        \
-    raise_lds((const void*)score_forest_v4<KT, RPT, RL>, lds);                \
-    hipLaunchKernelGGL((score_forest_v4<KT, RPT, RL>), dim3(blocks),          \
+    raise_lds((const void*)score_forest_v4<KT, RPT, RL, TI>, lds);            \
+    hipLaunchKernelGGL((score_forest_v4<KT, RPT, RL, TI>), dim3(blocks),      \
                        dim3(256), lds, stream, (const KT*)X,                  \
                        (const int2*)nodes, ncount, out, N, d, dpad, Tpad,     \
                        max_nodes, height_limit, fT, c_norm, finalize);        \
   } while (0)
+#define LS_ILP(KT, RPT, RL)                                                   \
+  do {                                                                        \
+    if (ilp == 8) LS(KT, RPT, RL, 8); else LS(KT, RPT, RL, 4);                \
+  } while (0)
   if (bf16) {
-    if (rows_lds && rpt == 2) LS(uint16_t, 2, true);
-    else if (rows_lds) LS(uint16_t, 1, true);
-    else LS(uint16_t, 1, false);
+    if (rows_lds && rpt == 2) LS_ILP(uint16_t, 2, true);
+    else if (rows_lds) LS_ILP(uint16_t, 1, true);
+    else LS_ILP(uint16_t, 1, false);
   } else {
-    if (rows_lds && rpt == 2) LS(uint32_t, 2, true);
-    else if (rows_lds) LS(uint32_t, 1, true);
-    else LS(uint32_t, 1, false);
+    if (rows_lds && rpt == 2) LS_ILP(uint32_t, 2, true);
+    else if (rows_lds) LS_ILP(uint32_t, 1, true);
+    else LS_ILP(uint32_t, 1, false);
   }
+#undef LS_ILP
 #undef LS
 }
 
