@@ -778,46 +778,20 @@ __device__ __forceinline__ float load_row_f32<uint32_t>(const uint32_t* X,
   return cv.f;
 }
 
-// Row storage for the dense-EIF walk: bf16 rows stay PACKED (2 per VGPR,
-// bf16->f32 unpack is an exact shift/mask), f32 rows are stored directly.
+// Row storage for the dense-EIF walk: rows are converted to f32 ONCE at
+// load (bf16->f32 is an exact shift) and held in registers across the
+// whole tree loop, so the dot needs zero per-visit unpack VALU. Register
+// budget: D=32 runs RPT=1 (32 row VGPRs), D<=16 runs RPT=2.
 template <typename KT, int D>
-struct RowReg;
-
-template <int D>
-struct RowReg<uint16_t, D> {
-  uint32_t pk[D / 2];
-  __device__ __forceinline__ void load(const uint16_t* X, int64_t base,
-                                       int d, bool ok) {
-#pragma unroll
-    for (int p = 0; p < D / 2; ++p) {
-      const uint32_t lo =
-          (ok && 2 * p < d) ? (uint32_t)X[base + 2 * p] : 0u;
-      const uint32_t hi =
-          (ok && 2 * p + 1 < d) ? (uint32_t)X[base + 2 * p + 1] : 0u;
-      pk[p] = lo | (hi << 16);
-    }
-  }
-  __device__ __forceinline__ float get(int j) const {  // j compile-time
-    union { uint32_t u; float f; } cv;
-    cv.u = (j & 1) ? (pk[j / 2] & 0xFFFF0000u) : (pk[j / 2] << 16);
-    return cv.f;
-  }
-};
-
-template <int D>
-struct RowReg<uint32_t, D> {
-  uint32_t pk[D];
-  __device__ __forceinline__ void load(const uint32_t* X, int64_t base,
-                                       int d, bool ok) {
+struct RowReg {
+  float v[D];
+  __device__ __forceinline__ void load(const KT* X, int64_t base, int d,
+                                       bool ok) {
 #pragma unroll
     for (int j = 0; j < D; ++j)
-      pk[j] = (ok && j < d) ? X[base + j] : 0u;
+      v[j] = (ok && j < d) ? load_row_f32<KT>(X, base + j) : 0.f;
   }
-  __device__ __forceinline__ float get(int j) const {
-    union { uint32_t u; float f; } cv;
-    cv.u = pk[j];
-    return cv.f;
-  }
+  __device__ __forceinline__ float get(int j) const { return v[j]; }
 };
 
 #define EIFD_THREADS 512
