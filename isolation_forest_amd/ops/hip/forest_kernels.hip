@@ -853,46 +853,40 @@ __global__ void __launch_bounds__(EIFD_THREADS, 4) score_extended_dense_v2(
       for (int r = 0; r < RPT; ++r) cur[r] = 0;
 
       for (int it = 0; it < height_limit; ++it) {
-        // phase 1: batched node reads; derive both successors up front so
-        // the select stays branchless (no exec-mask divergence)
         int2 nd[RPT];
 #pragma unroll
         for (int r = 0; r < RPT; ++r) nd[r] = tlds[cur[r]];
-        // phases 2+3: weight reads in chunks to bound live registers
-        // (quarter chunks at D=32 keep the kernel under 128 VGPRs)
+        // weight reads in chunks to bound live registers (quarter chunks at
+        // D=32 keep the kernel under 128 VGPRs). Lanes whose walk already
+        // ended (leaf self-loop: right == own id) are exec-masked out of
+        // the dot — masked lanes issue no LDS accesses, cutting the bank-
+        // conflict cycles that bound this kernel.
         constexpr int CHUNKS = (D >= 32) ? 4 : 2;
         constexpr int CW4 = D / (4 * CHUNKS);  // float4s per chunk
-        float a0[RPT], a1[RPT], a2[RPT], a3[RPT];
-#pragma unroll
-        for (int r = 0; r < RPT; ++r) a0[r] = a1[r] = a2[r] = a3[r] = 0.f;
-#pragma unroll
-        for (int ch = 0; ch < CHUNKS; ++ch) {
-          float4 w[RPT][CW4];
-#pragma unroll
-          for (int r = 0; r < RPT; ++r) {
-            const float4* wp = wlds + cur[r] * DW4 + ch * CW4;
-#pragma unroll
-            for (int j4 = 0; j4 < CW4; ++j4) w[r][j4] = wp[j4];
-          }
-#pragma unroll
-          for (int r = 0; r < RPT; ++r) {
-#pragma unroll
-            for (int j4 = 0; j4 < CW4; ++j4) {
-              const int j = ch * (D / CHUNKS) + 4 * j4;
-              a0[r] = __builtin_fmaf(w[r][j4].x, row[r].get(j + 0), a0[r]);
-              a1[r] = __builtin_fmaf(w[r][j4].y, row[r].get(j + 1), a1[r]);
-              a2[r] = __builtin_fmaf(w[r][j4].z, row[r].get(j + 2), a2[r]);
-              a3[r] = __builtin_fmaf(w[r][j4].w, row[r].get(j + 3), a3[r]);
-            }
-          }
-        }
 #pragma unroll
         for (int r = 0; r < RPT; ++r) {
-          const float dot = __fadd_rn(__fadd_rn(a0[r], a1[r]),
-                                      __fadd_rn(a2[r], a3[r]));
           const int right = pn_right(nd[r].x);
-          const int left = cur[r] + 1;
-          cur[r] = (dot < __int_as_float(nd[r].y)) ? left : right;
+          if (right != cur[r]) {
+            float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+#pragma unroll
+            for (int ch = 0; ch < CHUNKS; ++ch) {
+              float4 w[CW4];
+              const float4* wp = wlds + cur[r] * DW4 + ch * CW4;
+#pragma unroll
+              for (int j4 = 0; j4 < CW4; ++j4) w[j4] = wp[j4];
+#pragma unroll
+              for (int j4 = 0; j4 < CW4; ++j4) {
+                const int j = ch * (D / CHUNKS) + 4 * j4;
+                a0 = __builtin_fmaf(w[j4].x, row[r].get(j + 0), a0);
+                a1 = __builtin_fmaf(w[j4].y, row[r].get(j + 1), a1);
+                a2 = __builtin_fmaf(w[j4].z, row[r].get(j + 2), a2);
+                a3 = __builtin_fmaf(w[j4].w, row[r].get(j + 3), a3);
+              }
+            }
+            const float dot = __fadd_rn(__fadd_rn(a0, a1),
+                                        __fadd_rn(a2, a3));
+            cur[r] = (dot < __int_as_float(nd[r].y)) ? cur[r] + 1 : right;
+          }
         }
       }
 #pragma unroll
