@@ -217,6 +217,91 @@ def _compress(codec: str, block: bytes) -> bytes:
 
 
 # ---------------------------------------------------------------------------
+# pre-encoded container write / raw read (vectorized path, avro_fast.py)
+# ---------------------------------------------------------------------------
+
+
+def write_container_encoded(path: str, schema: dict, payload, reclens,
+                            codec: str = "deflate", extra_meta: dict = None,
+                            block_records: int = 65536, sync: bytes = None):
+    """Write a container from pre-encoded record bytes (numpy uint8) plus
+    per-record byte lengths; records are split into blocks on record
+    boundaries. ``extra_meta``: additional header metadata entries
+    (e.g. avro_fast.RECLENS_KEY) — other Avro readers ignore unknown keys."""
+    import numpy as _np
+
+    sync = sync or os.urandom(16)
+    schema_bytes = json.dumps(schema, separators=(",", ":")).encode()
+    out = io.BytesIO()
+    out.write(MAGIC)
+    meta = {"avro.schema": schema_bytes, "avro.codec": codec.encode()}
+    for k, v in (extra_meta or {}).items():
+        meta[k] = v
+    out.write(zigzag_encode(len(meta)))
+    for k, v in meta.items():
+        kb = k.encode()
+        out.write(zigzag_encode(len(kb)))
+        out.write(kb)
+        out.write(zigzag_encode(len(v)))
+        out.write(v)
+    out.write(zigzag_encode(0))
+    out.write(sync)
+
+    payload = _np.ascontiguousarray(payload, dtype=_np.uint8)
+    reclens = _np.ascontiguousarray(reclens, dtype=_np.int64)
+    nrec = len(reclens)
+    bounds = _np.zeros(nrec + 1, dtype=_np.int64)
+    _np.cumsum(reclens, out=bounds[1:])
+    i = 0
+    while i < nrec:
+        j = min(i + block_records, nrec)
+        raw = payload[bounds[i]:bounds[j]].tobytes()
+        block = _compress(codec, raw)
+        out.write(zigzag_encode(j - i))
+        out.write(zigzag_encode(len(block)))
+        out.write(block)
+        out.write(sync)
+        i = j
+    with open(path, "wb") as f:
+        f.write(out.getvalue())
+    return path
+
+
+def read_container_raw(path: str):
+    """Returns (schema_dict, meta_dict, payload_bytes, record_count) without
+    decoding records — the caller decodes (avro_fast) or falls back."""
+    data = open(path, "rb").read()
+    if data[:4] != MAGIC:
+        raise ValueError(f"{path} is not an Avro object container file")
+    r = Reader(data)
+    r.pos = 4
+    meta = {}
+    while True:
+        cnt = r.read_long()
+        if cnt == 0:
+            break
+        if cnt < 0:
+            r.read_long()
+            cnt = -cnt
+        for _ in range(cnt):
+            k = r.read_bytes(r.read_long()).decode()
+            meta[k] = r.read_bytes(r.read_long())
+    sync = r.read_bytes(16)
+    schema = json.loads(meta["avro.schema"].decode())
+    codec = meta.get("avro.codec", b"null").decode()
+    chunks = []
+    total = 0
+    while not r.at_end():
+        count = r.read_long()
+        size = r.read_long()
+        chunks.append(_decompress(codec, r.read_bytes(size)))
+        if r.read_bytes(16) != sync:
+            raise ValueError("avro sync marker mismatch")
+        total += count
+    return schema, meta, b"".join(chunks), total
+
+
+# ---------------------------------------------------------------------------
 # container read
 # ---------------------------------------------------------------------------
 

@@ -149,6 +149,72 @@ def extended_node_records(fr: ExtendedForest):
             yield {"treeID": t, "extendedNodeData": node}
 
 
+def _flat_live(forest):
+    """Row-major (tree asc, id asc) selection of live node slots — the same
+    record order the generators yield. Returns (tree_col, id_col, flatidx)."""
+    T, mn = forest.feature.shape
+    nc = forest.node_count.astype(np.int64)
+    live = np.arange(mn, dtype=np.int64)[None, :] < nc[:, None]
+    flat = np.nonzero(live.reshape(-1))[0]
+    tree_col = flat // mn
+    id_col = flat % mn
+    return tree_col, id_col, flat
+
+
+def _fast_save_standard(path: str, forest, codec: str):
+    from . import avro_fast
+
+    tree_col, id_col, flat = _flat_live(forest)
+    feat = forest.feature.reshape(-1)[flat].astype(np.int64)
+    leaf = feat == Forest.LEAF
+    right = forest.right.reshape(-1)[flat].astype(np.int64)
+    v64 = (forest.value64 if forest.value64 is not None else
+           forest.value.astype(np.float64)).reshape(-1)[flat]
+    ni = forest.num_instances.reshape(-1)[flat].astype(np.int64)
+    recs = {
+        "treeID": tree_col,
+        "id": id_col,
+        "leftChild": np.where(leaf, -1, id_col + 1),
+        "rightChild": np.where(leaf, -1, right),
+        "splitAttribute": np.where(leaf, -1, feat),
+        "splitValue": np.where(leaf, 0.0, v64),
+        "numInstances": np.where(leaf, ni, -1),
+    }
+    payload, reclens = avro_fast.encode_standard(recs)
+    avro_io.write_container_encoded(
+        path, avro_io.STANDARD_SCHEMA, payload, reclens, codec=codec,
+        extra_meta={avro_fast.RECLENS_KEY: avro_fast.pack_reclens(reclens)},
+    )
+
+
+def _fast_save_extended(path: str, fr, codec: str):
+    from . import avro_fast
+
+    tree_col, id_col, flat = _flat_live(fr)
+    feat = fr.feature.reshape(-1)[flat].astype(np.int64)
+    leaf = feat == ExtendedForest.LEAF
+    right = fr.right.reshape(-1)[flat].astype(np.int64)
+    off = fr.offset64.reshape(-1)[flat]
+    ni = fr.num_instances.reshape(-1)[flat].astype(np.int64)
+    counts = np.where(leaf, 0, feat)
+    nnz = fr.hyper_idx.shape[2]
+    indices = fr.hyper_idx.reshape(-1, nnz)[flat].astype(np.int64)
+    weights = fr.hyper_w.reshape(-1, nnz)[flat].astype(np.float32)
+    recs = {
+        "treeID": tree_col,
+        "id": id_col,
+        "leftChild": np.where(leaf, -1, id_col + 1),
+        "rightChild": np.where(leaf, -1, right),
+        "offset": np.where(leaf, 0.0, off),
+        "numInstances": np.where(leaf, ni, -1),
+    }
+    payload, reclens = avro_fast.encode_extended(recs, counts, indices, weights)
+    avro_io.write_container_encoded(
+        path, avro_io.EXTENDED_SCHEMA, payload, reclens, codec=codec,
+        extra_meta={avro_fast.RECLENS_KEY: avro_fast.pack_reclens(reclens)},
+    )
+
+
 def save_model(model, path: str, overwrite: bool = False, codec: str = "deflate"):
     _prep_dir(path, overwrite)
     forest = model.forest
@@ -167,20 +233,11 @@ def save_model(model, path: str, overwrite: bool = False, codec: str = "deflate"
     _write_metadata(path, meta)
     os.makedirs(os.path.join(path, "data"), exist_ok=True)
     fname = f"part-00000-{uuid.uuid4()}-c000.avro"
+    fpath = os.path.join(path, "data", fname)
     if extended:
-        avro_io.write_container(
-            os.path.join(path, "data", fname),
-            avro_io.EXTENDED_SCHEMA,
-            extended_node_records(forest),
-            codec=codec,
-        )
+        _fast_save_extended(fpath, forest, codec)
     else:
-        avro_io.write_container(
-            os.path.join(path, "data", fname),
-            avro_io.STANDARD_SCHEMA,
-            standard_node_records(forest),
-            codec=codec,
-        )
+        _fast_save_standard(fpath, forest, codec)
     open(os.path.join(path, "data", "_SUCCESS"), "w").close()
 
 
@@ -203,6 +260,125 @@ class WriteHandle:
 # ---------------------------------------------------------------------------
 # load
 # ---------------------------------------------------------------------------
+
+
+def _try_fast_columns(path: str, extended: bool):
+    """Vectorized decode of all data files via the ifa.reclens metadata;
+    returns a concatenated column dict or None (fall back to generic)."""
+    from . import avro_fast
+
+    files = sorted(glob.glob(os.path.join(path, "data", "*.avro")))
+    if not files:
+        raise FileNotFoundError(f"no avro data files under {path}/data")
+    parts = []
+    for f in files:
+        try:
+            schema, meta, payload, nrec = avro_io.read_container_raw(f)
+        except Exception:
+            return None
+        if avro_fast.RECLENS_KEY not in meta:
+            return None
+        reclens = avro_fast.unpack_reclens(meta[avro_fast.RECLENS_KEY])
+        if len(reclens) != nrec:
+            return None
+        starts = np.zeros(nrec, dtype=np.int64)
+        np.cumsum(reclens[:-1], out=starts[1:])
+        buf = np.frombuffer(payload, dtype=np.uint8)
+        try:
+            cols = (avro_fast.decode_extended(buf, starts) if extended
+                    else avro_fast.decode_standard(buf, starts))
+        except Exception:
+            return None
+        parts.append(cols)
+    out = {}
+    for k in parts[0]:
+        arrs = [p[k] for p in parts]
+        if arrs[0].ndim == 2:
+            w = max(a.shape[1] for a in arrs)
+            arrs = [np.pad(a, [(0, 0), (0, w - a.shape[1])]) for a in arrs]
+        out[k] = np.concatenate(arrs)
+    # fast path requires records grouped by tree with ascending pre-order
+    # ids (our writer's order); otherwise defer to the generic loader
+    tid, nid = out["treeID"], out["id"]
+    if len(tid) and not (
+        np.all(np.diff(tid) >= 0)
+        and np.all((np.diff(nid) == 1) | (np.diff(tid) > 0))
+        and np.all(nid[np.concatenate([[True], np.diff(tid) > 0])] == 0)
+    ):
+        return None
+    return out
+
+
+def _forest_from_columns(c, num_samples, num_features, total_num_features) -> Forest:
+    tid = c["treeID"].astype(np.int64)
+    nid = c["id"].astype(np.int64)
+    T = int(tid.max()) + 1 if len(tid) else 0
+    nc = np.bincount(tid, minlength=max(T, 1)).astype(np.int32)
+    max_nodes = int(nc.max(initial=1))
+    forest = empty_forest(T, max_nodes, num_samples, num_features,
+                          total_num_features)
+    if T == 0:
+        return forest
+    forest.node_count[:] = nc
+    leaf = c["leftChild"] == -1
+    internal = ~leaf
+    if not np.all(c["leftChild"][internal] == nid[internal] + 1):
+        bad = np.nonzero(c["leftChild"][internal] != nid[internal] + 1)[0][0]
+        t, i = tid[internal][bad], nid[internal][bad]
+        raise ValueError(
+            f"tree {t} node {i}: leftChild {c['leftChild'][internal][bad]} "
+            "breaks pre-order invariant (expected id+1)"
+        )
+    ti = (tid, nid)
+    forest.feature[ti] = np.where(leaf, Forest.LEAF,
+                                  c["splitAttribute"]).astype(np.int32)
+    forest.num_instances[ti] = np.where(leaf, c["numInstances"], -1)
+    forest.value[ti] = np.where(
+        leaf, avg_path_length(np.maximum(c["numInstances"], 0)),
+        c["splitValue"].astype(np.float32),
+    ).astype(np.float32)
+    forest.value64[ti] = np.where(leaf, 0.0, c["splitValue"])
+    forest.right[ti] = np.where(leaf, -1, c["rightChild"]).astype(np.int32)
+    return forest
+
+
+def _extended_forest_from_columns(c, num_samples, num_features,
+                                  total_num_features, params) -> ExtendedForest:
+    tid = c["treeID"].astype(np.int64)
+    nid = c["id"].astype(np.int64)
+    T = int(tid.max()) + 1 if len(tid) else 0
+    nc = np.bincount(tid, minlength=max(T, 1)).astype(np.int32)
+    max_nodes = int(nc.max(initial=1))
+    counts = c["_counts"].astype(np.int64)
+    nnz = int(counts.max(initial=0))
+    if params.is_set("extensionLevel"):
+        ext = params.get("extensionLevel")
+    else:
+        ext = max(nnz - 1, 0)
+    nnz = max(nnz, 1)
+    fr = empty_extended_forest(
+        T, max_nodes, nnz, num_samples, num_features, total_num_features, ext
+    )
+    if T == 0:
+        return fr
+    fr.node_count[:] = nc
+    leaf = c["leftChild"] == -1
+    if not np.all(c["leftChild"][~leaf] == nid[~leaf] + 1):
+        raise ValueError("pre-order invariant broken (leftChild != id+1)")
+    ti = (tid, nid)
+    fr.feature[ti] = np.where(leaf, ExtendedForest.LEAF,
+                              counts).astype(np.int32)
+    fr.num_instances[ti] = np.where(leaf, c["numInstances"], -1)
+    fr.value[ti] = np.where(
+        leaf, avg_path_length(np.maximum(c["numInstances"], 0)),
+        c["offset"].astype(np.float32),
+    ).astype(np.float32)
+    fr.offset64[ti] = np.where(leaf, 0.0, c["offset"])
+    fr.right[ti] = np.where(leaf, -1, c["rightChild"]).astype(np.int32)
+    w = c["_indices"].shape[1]
+    fr.hyper_idx[tid, nid, :w] = c["_indices"].astype(np.int32)
+    fr.hyper_w[tid, nid, :w] = c["_weights"].astype(np.float32)
+    return fr
 
 
 def _records_by_tree(path: str, field: str):
@@ -240,21 +416,32 @@ def load_model(path: str, expect_extended: bool):
     total_num_features = int(meta.get("totalNumFeatures", -1))
     threshold = float(meta.get("outlierScoreThreshold", -1.0))
 
+    cols = _try_fast_columns(path, extended=is_extended)
     if is_extended:
-        trees = _records_by_tree(path, "extendedNodeData")
-        forest = _extended_forest_from_records(
-            trees, num_samples, num_features, total_num_features, params
-        )
+        if cols is not None:
+            forest = _extended_forest_from_columns(
+                cols, num_samples, num_features, total_num_features, params
+            )
+        else:
+            trees = _records_by_tree(path, "extendedNodeData")
+            forest = _extended_forest_from_records(
+                trees, num_samples, num_features, total_num_features, params
+            )
         model = ExtendedIsolationForestModel(
             uid=meta.get("uid", "extended-isolation-forest_loaded"),
             forest=forest,
             params=params,
         )
     else:
-        trees = _records_by_tree(path, "nodeData")
-        forest = _forest_from_records(
-            trees, num_samples, num_features, total_num_features
-        )
+        if cols is not None:
+            forest = _forest_from_columns(
+                cols, num_samples, num_features, total_num_features
+            )
+        else:
+            trees = _records_by_tree(path, "nodeData")
+            forest = _forest_from_records(
+                trees, num_samples, num_features, total_num_features
+            )
         model = IsolationForestModel(
             uid=meta.get("uid", "isolation-forest_loaded"),
             forest=forest,

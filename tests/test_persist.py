@@ -219,3 +219,54 @@ class TestReferenceGoldenFixtures:
         for t in range(model.forest.num_trees):
             assert reloaded.forest.tree_to_string(t) == model.forest.tree_to_string(t)
         assert reloaded.outlier_score_threshold == model.outlier_score_threshold
+
+
+class TestFastCodec:
+    """avro_fast must emit byte-identical record streams to the generic
+    writer and decode them back exactly (fallback guard for files without
+    the ifa.reclens metadata is exercised by the reference-fixture tests)."""
+
+    def test_standard_stream_matches_generic(self, trained, tmp_path):
+        model, _ = trained
+        from isolation_forest_amd.persist import avro_io, model_io
+
+        p_generic = str(tmp_path / "g.avro")
+        avro_io.write_container(
+            p_generic, avro_io.STANDARD_SCHEMA,
+            model_io.standard_node_records(model.forest), codec="null",
+        )
+        p_fast = str(tmp_path / "f")
+        model_io.save_model(model, p_fast, codec="null")
+        f_fast = glob.glob(os.path.join(p_fast, "data", "*.avro"))[0]
+        _, _, payload_fast, n_fast = avro_io.read_container_raw(f_fast)
+        _, _, payload_gen, n_gen = avro_io.read_container_raw(p_generic)
+        assert n_fast == n_gen
+        assert payload_fast == payload_gen
+
+    def test_extended_stream_matches_generic(self, trained_ext, tmp_path):
+        model, _ = trained_ext
+        from isolation_forest_amd.persist import avro_io, model_io
+
+        p_generic = str(tmp_path / "g.avro")
+        avro_io.write_container(
+            p_generic, avro_io.EXTENDED_SCHEMA,
+            model_io.extended_node_records(model.forest), codec="null",
+        )
+        p_fast = str(tmp_path / "f")
+        model_io.save_model(model, p_fast, codec="null")
+        f_fast = glob.glob(os.path.join(p_fast, "data", "*.avro"))[0]
+        _, _, payload_fast, n_fast = avro_io.read_container_raw(f_fast)
+        _, _, payload_gen, n_gen = avro_io.read_container_raw(p_generic)
+        assert n_fast == n_gen
+        assert payload_fast == payload_gen
+
+    def test_generic_reader_accepts_fast_files(self, trained, tmp_path):
+        model, _ = trained
+        from isolation_forest_amd.persist import avro_io
+
+        p = str(tmp_path / "m")
+        model.save(p)
+        f = glob.glob(os.path.join(p, "data", "*.avro"))[0]
+        _, records = avro_io.read_container(f)
+        assert len(records) == int(model.forest.node_count.sum())
+        assert records[0]["nodeData"]["id"] == 0
