@@ -155,27 +155,54 @@ def _key32(vals: np.ndarray) -> np.ndarray:
     return k
 
 
-def _bf16_threshold_keys(split_vals: np.ndarray) -> np.ndarray:
-    """Smallest producible bf16 key whose value is >= the f32 split value,
-    shifted into the high 16 bits (kernel compares widened keys)."""
+def _bf16_threshold_keys_table(split_vals: np.ndarray) -> np.ndarray:
+    """Table/searchsorted reference implementation (used by tests to
+    cross-check the closed form below)."""
     vals, keys = _bf16_key_tables()
     idx = np.searchsorted(vals, split_vals.astype(np.float32), side="left")
     return keys[idx].astype(np.uint32) << np.uint32(16)
 
 
+def _bf16_threshold_keys(split_vals: np.ndarray) -> np.ndarray:
+    """Smallest producible bf16 key whose value is >= the f32 split value,
+    shifted into the high 16 bits (kernel compares widened keys).
+
+    Closed form: for s > 0 the smallest bf16 >= s is the truncated top-16
+    bits, +1 when any mantissa bits were cut (carry into the exponent /
+    into +inf is exactly right); for s <= 0 truncation already rounds
+    toward +inf. Then apply the device key16() transform."""
+    bits = np.ascontiguousarray(split_vals, dtype=np.float32).view(np.uint32)
+    neg = (bits & np.uint32(0x80000000)) != 0
+    b = (bits >> np.uint32(16)).astype(np.uint32)
+    frac = (bits & np.uint32(0xFFFF)) != 0
+    b = np.where(~neg & frac, b + np.uint32(1), b).astype(np.uint32)
+    # device key16(): +-0 -> 0x8000, monotone elsewhere (s is never NaN)
+    m = b & np.uint32(0x7FFF)
+    k = np.where(b & np.uint32(0x8000),
+                 np.uint32(0x7FFF) - m,
+                 np.uint32(0x8000) + m).astype(np.uint32)
+    k[m == 0] = np.uint32(0x8000)
+    return k << np.uint32(16)
+
+
 def _node_depths(feature: np.ndarray, right: np.ndarray) -> np.ndarray:
-    """Per-node depth via one pre-order sweep (parents precede children)."""
+    """Per-node depth via a level-synchronous frontier sweep (one vectorized
+    gather/scatter per level, ~height iterations total)."""
     T, mn = feature.shape
     depth = np.zeros((T, mn), dtype=np.int32)
-    rows = np.arange(T)
-    for i in range(mn - 1):
-        internal = feature[:, i] >= 0
-        if not internal.any():
-            continue
-        d1 = depth[:, i] + 1
-        depth[internal, i + 1] = d1[internal]
-        ri = rows[internal]
-        depth[ri, right[internal, i]] = d1[internal]
+    ft = np.repeat(np.arange(T, dtype=np.int64), 1)
+    fn = np.zeros(T, dtype=np.int64)  # frontier: root of every tree
+    level = 0
+    while len(ft):
+        internal = feature[ft, fn] >= 0
+        ft, fn = ft[internal], fn[internal]
+        if not len(ft):
+            break
+        level += 1
+        rt = right[ft, fn].astype(np.int64)
+        ft = np.concatenate([ft, ft])
+        fn = np.concatenate([fn + 1, rt])
+        depth[ft, fn] = level
     return depth
 
 

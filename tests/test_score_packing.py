@@ -175,3 +175,26 @@ class TestPackedV4:
             total = (total + w1[t, cur].view(np.float32)).astype(np.float32)
         np.testing.assert_array_equal(total.view(np.int32),
                                       oracle.view(np.int32))
+
+
+class TestThresholdKeyClosedForm:
+    def test_matches_table_reference(self):
+        from isolation_forest_amd.ops.gpu_engine import (
+            _bf16_threshold_keys, _bf16_threshold_keys_table)
+
+        rs = np.random.RandomState(3)
+        s = np.concatenate([
+            rs.normal(size=5000).astype(np.float32),
+            rs.normal(scale=1e-40, size=500).astype(np.float32),
+            rs.normal(scale=1e38, size=500).astype(np.float32),
+            # exact bf16 values and their neighbours
+            ((rs.randint(0, 1 << 16, size=2000).astype(np.uint32) << 16)
+             .view(np.float32)),
+            np.array([0.0, -0.0, 1e-45, -1e-45,
+                      np.finfo(np.float32).max, -np.finfo(np.float32).max,
+                      np.inf, -np.inf], dtype=np.float32),
+        ])
+        s = s[np.isfinite(s) | np.isinf(s)]
+        s = s[~np.isnan(s)]
+        np.testing.assert_array_equal(
+            _bf16_threshold_keys(s), _bf16_threshold_keys_table(s))
