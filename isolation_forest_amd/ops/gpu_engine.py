@@ -43,7 +43,7 @@ def build_forest(X: torch.Tensor, bag_idx: np.ndarray, feat_sub: np.ndarray,
     bag_t = torch.from_numpy(np.ascontiguousarray(bag_idx)).to(device)
     fs_t = torch.from_numpy(np.ascontiguousarray(feat_sub, dtype=np.int32)).to(device)
     bags = ext.bag_gather(X.contiguous(), bag_t)
-    feat, value, right, count, ncount = ext.build_forest(
+    feat, value, right, count, ncount, depth = ext.build_forest(
         bags, fs_t, seed, tree_id_offset, _leaf_lut(n, device), max_nodes,
         height_limit,
     )
@@ -56,7 +56,7 @@ def build_forest(X: torch.Tensor, bag_idx: np.ndarray, feat_sub: np.ndarray,
     v_np = np.where(f_np >= 0, v_np, np.where(f_np == Forest.LEAF, v_np, 0.0)).astype(
         np.float32
     )
-    return Forest(
+    forest = Forest(
         feature=f_np,
         value=v_np,
         right=r_np,
@@ -67,6 +67,12 @@ def build_forest(X: torch.Tensor, bag_idx: np.ndarray, feat_sub: np.ndarray,
         total_num_features=rp.total_features,
         value64=np.where(f_np >= 0, v_np, 0.0).astype(np.float64),
     )
+    # raw device tensors for scoring-pack reuse (skips the host round trip)
+    forest._device_raw = {
+        "device": str(device), "feat": feat, "value": value, "right": right,
+        "ncount": ncount, "depth": depth,
+    }
+    return forest
 
 
 def build_extended_forest(X: torch.Tensor, bag_idx: np.ndarray,
@@ -81,7 +87,8 @@ def build_extended_forest(X: torch.Tensor, bag_idx: np.ndarray,
     bag_t = torch.from_numpy(np.ascontiguousarray(bag_idx)).to(device)
     fs_t = torch.from_numpy(np.ascontiguousarray(feat_sub, dtype=np.int32)).to(device)
     bags = ext.bag_gather(X.contiguous(), bag_t)
-    feat, value, right, count, ncount, hidx, hw, off64 = ext.build_extended_forest(
+    (feat, value, right, count, ncount, hidx, hw, off64,
+     depth) = ext.build_extended_forest(
         bags, fs_t, seed, tree_id_offset, _leaf_lut(n, device), nnz, max_nodes,
         height_limit,
     )
@@ -99,7 +106,7 @@ def build_extended_forest(X: torch.Tensor, bag_idx: np.ndarray,
     hidx_np[~internal] = 0
     hw_np[~internal] = 0.0
     off64_np[~internal] = 0.0
-    return ExtendedForest(
+    fr = ExtendedForest(
         feature=f_np,
         value=v_np,
         right=r_np,
@@ -113,6 +120,11 @@ def build_extended_forest(X: torch.Tensor, bag_idx: np.ndarray,
         total_num_features=rp.total_features,
         extension_level=rp.extension_level,
     )
+    fr._device_raw = {
+        "device": str(device), "feat": feat, "value": value, "right": right,
+        "ncount": ncount, "depth": depth, "hw": hw,
+    }
+    return fr
 
 
 # ---------------------------------------------------------------------------
@@ -306,6 +318,97 @@ def _nodes_packed(forest) -> np.ndarray:
     return packed
 
 
+_TWO32 = 1 << 32
+_TWO31 = 1 << 31
+
+
+def _to_i32(x64: torch.Tensor) -> torch.Tensor:
+    """int64 (holding u32 bit patterns) -> int32 with C-style wraparound."""
+    return torch.where(x64 >= _TWO31, x64 - _TWO32, x64).to(torch.int32)
+
+
+def _bf16_threshold_keys_dev(vals_f32: torch.Tensor) -> torch.Tensor:
+    """Device mirror of _bf16_threshold_keys: int64 holding (t16 << 16)."""
+    bits = vals_f32.view(torch.int32).to(torch.int64) & 0xFFFFFFFF
+    neg = (bits & 0x80000000) != 0
+    b = bits >> 16
+    frac = (bits & 0xFFFF) != 0
+    b = torch.where(~neg & frac, b + 1, b)
+    m = b & 0x7FFF
+    k = torch.where((b & 0x8000) != 0, 0x7FFF - m, 0x8000 + m)
+    k = torch.where(m == 0, torch.full_like(k, 0x8000), k)
+    return k << 16
+
+
+def _key32_dev(vals_f32: torch.Tensor) -> torch.Tensor:
+    """Device mirror of _key32: int64 holding u32 keys (inputs never NaN)."""
+    b = vals_f32.view(torch.int32).to(torch.int64) & 0xFFFFFFFF
+    m = b & 0x7FFFFFFF
+    k = torch.where((b & 0x80000000) != 0, 0x7FFFFFFF - m, 0x80000000 + m)
+    return torch.where(m == 0, torch.full_like(k, 0x80000000), k)
+
+
+def _nodes_packed_v4_device(raw, num_trees: int, d_sentinel: int, bf16: bool):
+    """Torch-side v4 packing from the raw build outputs still on device —
+    identical bit layout to _nodes_packed_v4, no host round trip."""
+    feat, value = raw["feat"], raw["value"]
+    right, ncount, depth = raw["right"], raw["ncount"], raw["depth"]
+    T, mn = feat.shape
+    dev = feat.device
+    ids = torch.arange(mn, dtype=torch.int32, device=dev).expand(T, mn)
+    live = ids < ncount.to(torch.int32).unsqueeze(1)
+    internal = live & (feat >= 0)
+    leaf = live & (feat == -1)
+    w0 = torch.where(
+        internal,
+        feat | (right << 12),
+        torch.tensor(d_sentinel, dtype=torch.int32, device=dev) | (ids << 12),
+    )
+    tkeys = (_bf16_threshold_keys_dev(value) if bf16 else _key32_dev(value))
+    leafbits = (depth.to(torch.float32) + value).view(torch.int32)
+    w1 = torch.where(internal, _to_i32(tkeys),
+                     torch.where(leaf, leafbits,
+                                 torch.zeros_like(leafbits)))
+    pad = (-T) % 8
+    if pad:
+        w0p = (torch.tensor(d_sentinel, dtype=torch.int32, device=dev)
+               | (ids[:1] << 12)).expand(pad, mn).contiguous()
+        w0 = torch.cat([w0, w0p])
+        w1 = torch.cat([w1, torch.zeros(pad, mn, dtype=torch.int32,
+                                        device=dev)])
+    packed = torch.stack([w0, w1], dim=2).contiguous()
+    ncount_p = torch.cat([
+        ncount.to(torch.int32),
+        torch.ones(pad, dtype=torch.int32, device=dev)])
+    max_depth = int(depth.masked_fill(~live, 0).max().item())
+    return packed, ncount_p, max(max_depth, 1)
+
+
+def _eif_dense_packed_device(raw, D: int):
+    """Torch-side mirror of _eif_dense_packed (+ weight padding to D)."""
+    feat, value = raw["feat"], raw["value"]
+    right, ncount, depth = raw["right"], raw["ncount"], raw["depth"]
+    hw = raw["hw"]
+    T, mn = feat.shape
+    dev = feat.device
+    ids = torch.arange(mn, dtype=torch.int32, device=dev).expand(T, mn)
+    live = ids < ncount.to(torch.int32).unsqueeze(1)
+    internal = live & (feat >= 0)
+    leaf = live & (feat == -1)
+    w0 = torch.where(internal, right, ids) << 12
+    ninf = torch.tensor(float("-inf"), device=dev)
+    w1 = torch.where(internal, value, ninf).view(torch.int32)
+    packed = torch.stack([w0, w1], dim=2).contiguous()
+    values = torch.where(leaf, depth.to(torch.float32) + value,
+                         torch.zeros_like(value)).contiguous()
+    d = hw.shape[2]
+    if d < D:
+        hw = torch.cat([hw, torch.zeros(T, mn, D - d, dtype=hw.dtype,
+                                        device=dev)], dim=2)
+    max_depth = int(depth.masked_fill(~live, 0).max().item())
+    return packed, values, hw.contiguous(), max(max_depth, 1)
+
+
 def _device_forest(model, device, v4_key=None):
     """Cached device copy of the packed forest. v4_key = (d, bf16) selects
     the standard-scoring v4 packing; None = the EIF packing."""
@@ -314,29 +417,44 @@ def _device_forest(model, device, v4_key=None):
     if key not in cache:
         forest = model.forest
         extra = {}
+        raw = getattr(forest, "_device_raw", None)
+        if raw is not None and raw["device"] != str(device):
+            raw = None
         if v4_key == "eif_dense":
-            packed, values, max_depth = _eif_dense_packed(forest)
-            aos = torch.from_numpy(packed).to(device)
-            ncount = torch.from_numpy(
-                np.ascontiguousarray(forest.node_count, dtype=np.int32)
-            ).to(device)
-            extra["values"] = torch.from_numpy(values).to(device)
-            # pad weights to the kernel's compile-time D columns
             d = forest.hyper_w.shape[2]
             D = 8 if d <= 8 else (16 if d <= 16 else 32)
-            hw = forest.hyper_w
-            if d < D:
-                T, mn, _ = hw.shape
-                hw = np.concatenate(
-                    [hw, np.zeros((T, mn, D - d), dtype=np.float32)], axis=2
-                )
-            extra["hw"] = torch.from_numpy(np.ascontiguousarray(hw)).to(device)
+            if raw is not None:
+                aos, values_t, hw_t, max_depth = _eif_dense_packed_device(
+                    raw, D)
+                ncount = raw["ncount"]
+                extra["values"] = values_t
+                extra["hw"] = hw_t
+            else:
+                packed, values, max_depth = _eif_dense_packed(forest)
+                aos = torch.from_numpy(packed).to(device)
+                ncount = torch.from_numpy(
+                    np.ascontiguousarray(forest.node_count, dtype=np.int32)
+                ).to(device)
+                extra["values"] = torch.from_numpy(values).to(device)
+                hw = forest.hyper_w
+                if d < D:
+                    T, mn, _ = hw.shape
+                    hw = np.concatenate(
+                        [hw, np.zeros((T, mn, D - d), dtype=np.float32)],
+                        axis=2,
+                    )
+                extra["hw"] = torch.from_numpy(
+                    np.ascontiguousarray(hw)).to(device)
             extra["height"] = max_depth
         elif v4_key is not None:
             d, bf16 = v4_key
-            packed, ncount_np, max_depth = _nodes_packed_v4(forest, d, bf16)
-            aos = torch.from_numpy(packed).to(device)
-            ncount = torch.from_numpy(ncount_np).to(device)
+            if raw is not None:
+                aos, ncount, max_depth = _nodes_packed_v4_device(
+                    raw, forest.num_trees, d, bf16)
+            else:
+                packed, ncount_np, max_depth = _nodes_packed_v4(forest, d, bf16)
+                aos = torch.from_numpy(packed).to(device)
+                ncount = torch.from_numpy(ncount_np).to(device)
             extra["height"] = max_depth
         else:
             aos = torch.from_numpy(_nodes_packed(forest)).to(device)

@@ -16,7 +16,7 @@ void launch_bag_gather(bool bf16, const void* X, const int64_t* bag_idx,
 
 void launch_build_forest(const float* bags, const int32_t* feat_sub,
                          int32_t* feat, float* value, int32_t* right,
-                         int32_t* count, int32_t* ncount,
+                         int32_t* count, int32_t* ncount, int32_t* depth,
                          const float* leaf_lut, uint64_t seed,
                          int32_t tree_id_offset, int32_t T, int32_t n,
                          int32_t d, int32_t k, int32_t max_nodes,
@@ -26,6 +26,7 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
                                   int32_t* feat, float* value, int32_t* right,
                                   int32_t* count, int32_t* ncount,
                                   int32_t* hidx, float* hw, double* off64,
+                                  int32_t* depth,
                                   const float* leaf_lut, uint64_t seed,
                                   int32_t tree_id_offset, int32_t T, int32_t n,
                                   int32_t d, int32_t k, int32_t nnz,
@@ -135,6 +136,7 @@ std::vector<torch::Tensor> build_forest(torch::Tensor bags,
   auto right = torch::full({T, max_nodes}, -1, opts);
   auto count = torch::empty({T, max_nodes}, opts);
   auto ncount = torch::empty({T}, opts);
+  auto depth = torch::zeros({T, max_nodes}, opts);
 
   size_t lds = align16(4 * n + 2 * k) + (height_limit + 8) * 8;
   TORCH_CHECK(lds <= kMaxLds, "build LDS request too large");
@@ -142,10 +144,11 @@ std::vector<torch::Tensor> build_forest(torch::Tensor bags,
       bags.data_ptr<float>(), feat_sub.data_ptr<int32_t>(),
       feat.data_ptr<int32_t>(), value.data_ptr<float>(),
       right.data_ptr<int32_t>(), count.data_ptr<int32_t>(),
-      ncount.data_ptr<int32_t>(), leaf_lut.data_ptr<float>(), (uint64_t)seed,
+      ncount.data_ptr<int32_t>(), depth.data_ptr<int32_t>(),
+      leaf_lut.data_ptr<float>(), (uint64_t)seed,
       (int32_t)tree_id_offset, (int32_t)T, (int32_t)n, (int32_t)d, (int32_t)k,
       (int32_t)max_nodes, (int32_t)height_limit, lds, current_stream());
-  return {feat, value, right, count, ncount};
+  return {feat, value, right, count, ncount, depth};
 }
 
 std::vector<torch::Tensor> build_extended_forest(
@@ -173,6 +176,7 @@ std::vector<torch::Tensor> build_extended_forest(
   auto hw = torch::zeros({T, max_nodes, nnz}, bags.options());
   auto off64 =
       torch::zeros({T, max_nodes}, bags.options().dtype(torch::kFloat64));
+  auto depth = torch::zeros({T, max_nodes}, opts);
 
   size_t lds =
       align16(4 * n + 2 * k) + align16(8 * nnz) + (height_limit + 8) * 8 + 32;
@@ -183,10 +187,11 @@ std::vector<torch::Tensor> build_extended_forest(
       right.data_ptr<int32_t>(), count.data_ptr<int32_t>(),
       ncount.data_ptr<int32_t>(), hidx.data_ptr<int32_t>(),
       hw.data_ptr<float>(), off64.data_ptr<double>(),
+      depth.data_ptr<int32_t>(),
       leaf_lut.data_ptr<float>(), (uint64_t)seed, (int32_t)tree_id_offset,
       (int32_t)T, (int32_t)n, (int32_t)d, (int32_t)k, (int32_t)nnz,
       (int32_t)max_nodes, (int32_t)height_limit, lds, current_stream());
-  return {feat, value, right, count, ncount, hidx, hw, off64};
+  return {feat, value, right, count, ncount, hidx, hw, off64, depth};
 }
 
 torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,

@@ -109,6 +109,7 @@ __global__ void __launch_bounds__(WAVE) build_forest_kernel(
     int32_t* __restrict__ out_right,       // [T][max_nodes]
     int32_t* __restrict__ out_count,       // [T][max_nodes] (leaf counts; -1)
     int32_t* __restrict__ out_ncount,      // [T]
+    int32_t* __restrict__ out_depth,       // [T][max_nodes]
     const float* __restrict__ leaf_lut,    // [n+1]: c(m) float32
     uint64_t seed, int32_t tree_id_offset, int32_t n, int32_t d, int32_t k,
     int32_t max_nodes, int32_t height_limit) {
@@ -126,6 +127,7 @@ __global__ void __launch_bounds__(WAVE) build_forest_kernel(
   float* value = out_value + (int64_t)t * max_nodes;
   int32_t* right = out_right + (int64_t)t * max_nodes;
   int32_t* count = out_count + (int64_t)t * max_nodes;
+  int32_t* depth = out_depth + (int64_t)t * max_nodes;
   const int32_t* features = feat_sub + (int64_t)t * k;
 
   for (int i = lane; i < n; i += WAVE) idx[i] = (uint16_t)i;
@@ -137,7 +139,10 @@ __global__ void __launch_bounds__(WAVE) build_forest_kernel(
   while (sp > 0) {
     SegEntry e = stack[--sp];
     const int node = next_id++;
-    if (e.patch >= 0 && lane == 0) right[e.patch] = node;
+    if (lane == 0) {
+      if (e.patch >= 0) right[e.patch] = node;
+      depth[node] = e.height;
+    }
     const int start = e.start, end = e.end, m = end - start;
 
     if (m <= 1 || e.height >= height_limit) {
@@ -249,6 +254,7 @@ __global__ void __launch_bounds__(WAVE) build_extended_forest_kernel(
     int32_t* __restrict__ out_hidx,    // [T][max_nodes][nnz]
     float* __restrict__ out_hw,        // [T][max_nodes][nnz]
     double* __restrict__ out_off64,    // [T][max_nodes]
+    int32_t* __restrict__ out_depth,   // [T][max_nodes]
     const float* __restrict__ leaf_lut, uint64_t seed, int32_t tree_id_offset,
     int32_t n, int32_t d, int32_t k, int32_t nnz, int32_t max_nodes,
     int32_t height_limit) {
@@ -272,6 +278,7 @@ __global__ void __launch_bounds__(WAVE) build_extended_forest_kernel(
   int32_t* hidx = out_hidx + (int64_t)t * max_nodes * nnz;
   float* hw = out_hw + (int64_t)t * max_nodes * nnz;
   double* off64p = out_off64 + (int64_t)t * max_nodes;
+  int32_t* depth = out_depth + (int64_t)t * max_nodes;
   const int32_t* features = feat_sub + (int64_t)t * k;
 
   for (int i = lane; i < n; i += WAVE) idx[i] = (uint16_t)i;
@@ -283,7 +290,10 @@ __global__ void __launch_bounds__(WAVE) build_extended_forest_kernel(
   while (sp > 0) {
     SegEntry e = stack[--sp];
     const int node = next_id++;
-    if (e.patch >= 0 && lane == 0) right[e.patch] = node;
+    if (lane == 0) {
+      if (e.patch >= 0) right[e.patch] = node;
+      depth[node] = e.height;
+    }
     const int start = e.start, end = e.end, m = end - start;
 
     if (m <= 1 || e.height >= height_limit) {
@@ -1033,7 +1043,7 @@ void launch_bag_gather(bool bf16, const void* X, const int64_t* bag_idx,
 
 void launch_build_forest(const float* bags, const int32_t* feat_sub,
                          int32_t* feat, float* value, int32_t* right,
-                         int32_t* count, int32_t* ncount,
+                         int32_t* count, int32_t* ncount, int32_t* depth,
                          const float* leaf_lut, uint64_t seed,
                          int32_t tree_id_offset, int32_t T, int32_t n,
                          int32_t d, int32_t k, int32_t max_nodes,
@@ -1042,14 +1052,15 @@ void launch_build_forest(const float* bags, const int32_t* feat_sub,
   raise_lds((const void*)build_forest_kernel, lds);
   hipLaunchKernelGGL(build_forest_kernel, dim3(T), dim3(WAVE), lds, stream,
                      bags, feat_sub, feat, value, right, count, ncount,
-                     leaf_lut, seed, tree_id_offset, n, d, k, max_nodes,
-                     height_limit);
+                     depth, leaf_lut, seed, tree_id_offset, n, d, k,
+                     max_nodes, height_limit);
 }
 
 void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
                                   int32_t* feat, float* value, int32_t* right,
                                   int32_t* count, int32_t* ncount,
                                   int32_t* hidx, float* hw, double* off64,
+                                  int32_t* depth,
                                   const float* leaf_lut, uint64_t seed,
                                   int32_t tree_id_offset, int32_t T, int32_t n,
                                   int32_t d, int32_t k, int32_t nnz,
@@ -1058,8 +1069,8 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
   raise_lds((const void*)build_extended_forest_kernel, lds);
   hipLaunchKernelGGL(build_extended_forest_kernel, dim3(T), dim3(WAVE), lds,
                      stream, bags, feat_sub, feat, value, right, count, ncount,
-                     hidx, hw, off64, leaf_lut, seed, tree_id_offset, n, d, k,
-                     nnz, max_nodes, height_limit);
+                     hidx, hw, off64, depth, leaf_lut, seed, tree_id_offset,
+                     n, d, k, nnz, max_nodes, height_limit);
 }
 
 void launch_score_forest(bool bf16, int rpt, bool rows_lds, int ilp,

@@ -230,3 +230,54 @@ class TestEndToEndGPU:
         s_gpu = model.score(Xt).cpu().numpy()
         s_cpu = loaded.score(torch.from_numpy(X)).numpy()
         np.testing.assert_allclose(s_cpu, s_gpu, rtol=0, atol=2e-7)
+
+
+class TestDevicePacking:
+    """The torch-side scoring pack (built from raw device build outputs)
+    must be BITWISE identical to the host numpy pack."""
+
+    def test_standard_pack_matches_host(self, dev):
+        from isolation_forest_amd.ops import gpu_engine
+
+        X = torch.from_numpy(make_data(20000, 12, seed=21)).to(dev)
+        model = IsolationForest(numEstimators=64, randomSeed=5).fit(X)
+        forest = model.forest
+        raw = forest._device_raw
+        for bf16 in (True, False):
+            aos_dev, nc_dev, h_dev = gpu_engine._nodes_packed_v4_device(
+                raw, forest.num_trees, 12, bf16)
+            aos_host, nc_host, h_host = gpu_engine._nodes_packed_v4(
+                forest, 12, bf16)
+            np.testing.assert_array_equal(aos_dev.cpu().numpy(), aos_host)
+            np.testing.assert_array_equal(nc_dev.cpu().numpy(), nc_host)
+            assert h_dev == h_host
+
+    def test_extended_pack_matches_host(self, dev):
+        from isolation_forest_amd.ops import gpu_engine
+
+        X = torch.from_numpy(make_data(15000, 6, seed=22)).to(dev)
+        model = ExtendedIsolationForest(numEstimators=32, randomSeed=7).fit(X)
+        forest = model.forest
+        raw = forest._device_raw
+        aos_dev, vals_dev, hw_dev, h_dev = gpu_engine._eif_dense_packed_device(
+            raw, 8)
+        aos_host, vals_host, h_host = gpu_engine._eif_dense_packed(forest)
+        np.testing.assert_array_equal(aos_dev.cpu().numpy(), aos_host)
+        np.testing.assert_array_equal(vals_dev.cpu().numpy(), vals_host)
+        assert h_dev == h_host
+        hw_host = forest.hyper_w
+        np.testing.assert_array_equal(
+            hw_dev.cpu().numpy()[:, :, :6], hw_host)
+        assert hw_dev.shape[2] == 8
+        assert float(np.abs(hw_dev.cpu().numpy()[:, :, 6:]).max()) == 0.0
+
+    def test_scores_identical_both_paths(self, dev):
+        X = torch.from_numpy(make_data(30000, 10, seed=23)).to(dev)
+        model = IsolationForest(numEstimators=64, randomSeed=9).fit(X)
+        s_dev_pack = model.score(X).cpu().numpy()
+        # force the host packing path
+        del model.forest._device_raw
+        model._gpu_forest_cache = {}
+        s_host_pack = model.score(X).cpu().numpy()
+        np.testing.assert_array_equal(
+            s_dev_pack.view(np.int32), s_host_pack.view(np.int32))
