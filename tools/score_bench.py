@@ -27,6 +27,7 @@ def main():
     ap.add_argument("--reps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--extended", action="store_true")
+    ap.add_argument("--extension-level", type=int, default=None)
     args = ap.parse_args()
 
     from isolation_forest_amd import ExtendedIsolationForest, IsolationForest
@@ -41,8 +42,11 @@ def main():
         X = X.to(torch.bfloat16)
 
     cls = ExtendedIsolationForest if args.extended else IsolationForest
+    kw = {}
+    if args.extended and args.extension_level is not None:
+        kw["extensionLevel"] = args.extension_level
     est = cls(numEstimators=args.trees, maxSamples=float(args.max_samples),
-              randomSeed=3)
+              randomSeed=3, **kw)
     t0 = time.perf_counter()
     model = est.fit(X)
     torch.cuda.synchronize()
@@ -58,6 +62,7 @@ def main():
     dt = (time.perf_counter() - t0) / args.reps
     print(json.dumps({
         "kernel": "score_extended" if args.extended else "score",
+        "ext_level": args.extension_level if args.extended else None,
         "rows": args.rows, "features": args.features, "trees": args.trees,
         "dtype": args.dtype, "fit_s": round(t_fit, 4),
         "score_ms": round(dt * 1e3, 3),
