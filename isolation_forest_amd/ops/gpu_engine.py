@@ -122,7 +122,7 @@ def build_extended_forest(X: torch.Tensor, bag_idx: np.ndarray,
     )
     fr._device_raw = {
         "device": str(device), "feat": feat, "value": value, "right": right,
-        "ncount": ncount, "depth": depth, "hw": hw,
+        "ncount": ncount, "depth": depth, "hidx": hidx, "hw": hw,
     }
     return fr
 
@@ -264,11 +264,26 @@ def _nodes_packed_v4(forest, d_sentinel: int, bf16: bool):
     return packed, ncount, max(max_depth, 1)
 
 
-def _eif_dense_packed(forest):
+def _densify_weights(hidx: np.ndarray, hw: np.ndarray, counts: np.ndarray,
+                     D: int) -> np.ndarray:
+    """Scatter sparse hyperplanes [T, mn, nnz] into dense rows [T, mn, D].
+    Padded slots (j >= counts) are redirected to a scratch column so their
+    zero index cannot clobber a real coordinate-0 weight."""
+    T, mn, nnz = hidx.shape
+    j = np.arange(nnz, dtype=np.int64)[None, None, :]
+    idx = np.where(j < counts[:, :, None], hidx.astype(np.int64), D)
+    dense = np.zeros((T, mn, D + 1), dtype=np.float32)
+    np.put_along_axis(dense, idx, hw.astype(np.float32), axis=2)
+    return np.ascontiguousarray(dense[:, :, :D])
+
+
+def _eif_dense_packed(forest, D: int):
     """Packing for score_extended_dense_v2: w0 = right<<12 (leaf/pad: own id
     -> self-loop), w1 = offset f32 (leaf/pad: -inf so the zero weight row's
-    dot=0 routes right), plus a depth-folded leaf-value array.
-    Returns (packed int32 [T, mn, 2], values f32 [T, mn], max_depth)."""
+    dot=0 routes right), plus a depth-folded leaf-value array and the
+    DENSIFIED weight matrix (sparse hyperplanes scattered to D columns).
+    Returns (packed int32 [T, mn, 2], values f32 [T, mn], hw_dense, max_depth).
+    """
     T, mn = forest.feature.shape
     if mn > 32767:
         raise ValueError("forest too deep for the packed node format")
@@ -291,9 +306,11 @@ def _eif_dense_packed(forest):
         depth.astype(np.float32) + forest.value.astype(np.float32),
         np.float32(0.0),
     ).astype(np.float32)
+    counts = np.where(internal, feat, 0).astype(np.int64)
+    hw_dense = _densify_weights(forest.hyper_idx, forest.hyper_w, counts, D)
     live = internal | leaf
     max_depth = int(depth[live].max()) if live.any() else 0
-    return packed, values, max(max_depth, 1)
+    return packed, values, hw_dense, max(max_depth, 1)
 
 
 def _nodes_packed(forest) -> np.ndarray:
@@ -385,10 +402,10 @@ def _nodes_packed_v4_device(raw, num_trees: int, d_sentinel: int, bf16: bool):
 
 
 def _eif_dense_packed_device(raw, D: int):
-    """Torch-side mirror of _eif_dense_packed (+ weight padding to D)."""
+    """Torch-side mirror of _eif_dense_packed (densified weights)."""
     feat, value = raw["feat"], raw["value"]
     right, ncount, depth = raw["right"], raw["ncount"], raw["depth"]
-    hw = raw["hw"]
+    hidx, hw = raw["hidx"], raw["hw"]
     T, mn = feat.shape
     dev = feat.device
     ids = torch.arange(mn, dtype=torch.int32, device=dev).expand(T, mn)
@@ -401,12 +418,16 @@ def _eif_dense_packed_device(raw, D: int):
     packed = torch.stack([w0, w1], dim=2).contiguous()
     values = torch.where(leaf, depth.to(torch.float32) + value,
                          torch.zeros_like(value)).contiguous()
-    d = hw.shape[2]
-    if d < D:
-        hw = torch.cat([hw, torch.zeros(T, mn, D - d, dtype=hw.dtype,
-                                        device=dev)], dim=2)
+    counts = torch.where(internal, feat, torch.zeros_like(feat)).to(torch.int64)
+    nnz = hidx.shape[2]
+    j = torch.arange(nnz, dtype=torch.int64, device=dev).view(1, 1, nnz)
+    idx = torch.where(j < counts.unsqueeze(2), hidx.to(torch.int64),
+                      torch.tensor(D, dtype=torch.int64, device=dev))
+    dense = torch.zeros(T, mn, D + 1, dtype=torch.float32, device=dev)
+    dense.scatter_(2, idx, hw)
+    hw_dense = dense[:, :, :D].contiguous()
     max_depth = int(depth.masked_fill(~live, 0).max().item())
-    return packed, values, hw.contiguous(), max(max_depth, 1)
+    return packed, values, hw_dense, max(max_depth, 1)
 
 
 def _device_forest(model, device, v4_key=None):
@@ -420,9 +441,8 @@ def _device_forest(model, device, v4_key=None):
         raw = getattr(forest, "_device_raw", None)
         if raw is not None and raw["device"] != str(device):
             raw = None
-        if v4_key == "eif_dense":
-            d = forest.hyper_w.shape[2]
-            D = 8 if d <= 8 else (16 if d <= 16 else 32)
+        if isinstance(v4_key, tuple) and v4_key[0] == "eif_dense":
+            D = v4_key[1]
             if raw is not None:
                 aos, values_t, hw_t, max_depth = _eif_dense_packed_device(
                     raw, D)
@@ -430,21 +450,14 @@ def _device_forest(model, device, v4_key=None):
                 extra["values"] = values_t
                 extra["hw"] = hw_t
             else:
-                packed, values, max_depth = _eif_dense_packed(forest)
+                packed, values, hw_dense, max_depth = _eif_dense_packed(
+                    forest, D)
                 aos = torch.from_numpy(packed).to(device)
                 ncount = torch.from_numpy(
                     np.ascontiguousarray(forest.node_count, dtype=np.int32)
                 ).to(device)
                 extra["values"] = torch.from_numpy(values).to(device)
-                hw = forest.hyper_w
-                if d < D:
-                    T, mn, _ = hw.shape
-                    hw = np.concatenate(
-                        [hw, np.zeros((T, mn, D - d), dtype=np.float32)],
-                        axis=2,
-                    )
-                extra["hw"] = torch.from_numpy(
-                    np.ascontiguousarray(hw)).to(device)
+                extra["hw"] = torch.from_numpy(hw_dense).to(device)
             extra["height"] = max_depth
         elif v4_key is not None:
             d, bf16 = v4_key
@@ -491,8 +504,15 @@ def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torc
     forest = model.forest
     c = float(avg_path_length(forest.num_samples))
     d = int(X.shape[1])
-    if forest.nnz == d and d <= 32:
-        aos, ncount, extra = _device_forest(model, X.device, v4_key="eif_dense")
+    nnz = forest.nnz
+    # densified-dense walk wins over the strict-order sparse kernel once
+    # hyperplanes carry more than ~6 coordinates (measured crossover,
+    # profiles/r01_bench_and_kernels.md); the sparse kernel stays for
+    # low extension levels and wide feature spaces.
+    if d <= 32 and (nnz == d or nnz >= 6):
+        D = 8 if d <= 8 else (16 if d <= 16 else 32)
+        aos, ncount, extra = _device_forest(
+            model, X.device, v4_key=("eif_dense", D))
         return ext.score_extended_dense_v2(
             X.contiguous(), aos, extra["values"], extra["hw"], ncount,
             extra["height"], c, finalize,
