@@ -244,6 +244,9 @@ torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
     lds = node_bytes;
     break;
   }
+  if (const char* e = getenv("IFA_SCORE_FORCE_GLOBAL")) {
+    if (atoi(e)) { rows_lds = false; rpt = 1; lds = node_bytes; }
+  }
   int64_t rows_per_block = (int64_t)(rows_lds ? rpt : 1) * 256;
   int blocks = (int)std::min<int64_t>(
       (N + rows_per_block - 1) / rows_per_block, 8192);
