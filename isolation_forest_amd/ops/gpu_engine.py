@@ -277,13 +277,10 @@ def _densify_weights(hidx: np.ndarray, hw: np.ndarray, counts: np.ndarray,
     return np.ascontiguousarray(dense[:, :, :D])
 
 
-def _eif_dense_packed(forest, D: int):
-    """Packing for score_extended_dense_v2: w0 = right<<12 (leaf/pad: own id
-    -> self-loop), w1 = offset f32 (leaf/pad: -inf so the zero weight row's
-    dot=0 routes right), plus a depth-folded leaf-value array and the
-    DENSIFIED weight matrix (sparse hyperplanes scattered to D columns).
-    Returns (packed int32 [T, mn, 2], values f32 [T, mn], hw_dense, max_depth).
-    """
+def _eif_nodes_values(forest):
+    """Shared EIF v2 node packing: w0 = right<<12 (leaf/pad: own id ->
+    self-loop), w1 = offset f32 (leaf/pad: -inf), depth-folded leaf values.
+    Returns (packed int32 [T, mn, 2], values f32 [T, mn], max_depth)."""
     T, mn = forest.feature.shape
     if mn > 32767:
         raise ValueError("forest too deep for the packed node format")
@@ -306,11 +303,19 @@ def _eif_dense_packed(forest, D: int):
         depth.astype(np.float32) + forest.value.astype(np.float32),
         np.float32(0.0),
     ).astype(np.float32)
-    counts = np.where(internal, feat, 0).astype(np.int64)
-    hw_dense = _densify_weights(forest.hyper_idx, forest.hyper_w, counts, D)
     live = internal | leaf
     max_depth = int(depth[live].max()) if live.any() else 0
-    return packed, values, hw_dense, max(max_depth, 1)
+    return packed, values, max(max_depth, 1)
+
+
+def _eif_dense_packed(forest, D: int):
+    """EIF v2 packing + DENSIFIED weight matrix (sparse hyperplanes
+    scattered to D columns) for score_extended_dense_v2."""
+    packed, values, max_depth = _eif_nodes_values(forest)
+    feat = forest.feature
+    counts = np.where(feat >= 0, feat, 0).astype(np.int64)
+    hw_dense = _densify_weights(forest.hyper_idx, forest.hyper_w, counts, D)
+    return packed, values, hw_dense, max_depth
 
 
 def _nodes_packed(forest) -> np.ndarray:
@@ -401,11 +406,10 @@ def _nodes_packed_v4_device(raw, num_trees: int, d_sentinel: int, bf16: bool):
     return packed, ncount_p, max(max_depth, 1)
 
 
-def _eif_dense_packed_device(raw, D: int):
-    """Torch-side mirror of _eif_dense_packed (densified weights)."""
+def _eif_nodes_values_device(raw):
+    """Torch-side mirror of _eif_nodes_values."""
     feat, value = raw["feat"], raw["value"]
     right, ncount, depth = raw["right"], raw["ncount"], raw["depth"]
-    hidx, hw = raw["hidx"], raw["hw"]
     T, mn = feat.shape
     dev = feat.device
     ids = torch.arange(mn, dtype=torch.int32, device=dev).expand(T, mn)
@@ -418,6 +422,16 @@ def _eif_dense_packed_device(raw, D: int):
     packed = torch.stack([w0, w1], dim=2).contiguous()
     values = torch.where(leaf, depth.to(torch.float32) + value,
                          torch.zeros_like(value)).contiguous()
+    max_depth = int(depth.masked_fill(~live, 0).max().item())
+    return packed, values, internal, max(max_depth, 1)
+
+
+def _eif_dense_packed_device(raw, D: int):
+    """Torch-side mirror of _eif_dense_packed (densified weights)."""
+    packed, values, internal, max_depth = _eif_nodes_values_device(raw)
+    feat, hidx, hw = raw["feat"], raw["hidx"], raw["hw"]
+    T, mn = feat.shape
+    dev = feat.device
     counts = torch.where(internal, feat, torch.zeros_like(feat)).to(torch.int64)
     nnz = hidx.shape[2]
     j = torch.arange(nnz, dtype=torch.int64, device=dev).view(1, 1, nnz)
@@ -426,8 +440,7 @@ def _eif_dense_packed_device(raw, D: int):
     dense = torch.zeros(T, mn, D + 1, dtype=torch.float32, device=dev)
     dense.scatter_(2, idx, hw)
     hw_dense = dense[:, :, :D].contiguous()
-    max_depth = int(depth.masked_fill(~live, 0).max().item())
-    return packed, values, hw_dense, max(max_depth, 1)
+    return packed, values, hw_dense, max_depth
 
 
 def _device_forest(model, device, v4_key=None):
@@ -441,7 +454,26 @@ def _device_forest(model, device, v4_key=None):
         raw = getattr(forest, "_device_raw", None)
         if raw is not None and raw["device"] != str(device):
             raw = None
-        if isinstance(v4_key, tuple) and v4_key[0] == "eif_dense":
+        if v4_key == "eif_sparse":
+            if raw is not None:
+                aos, values_t, _, max_depth = _eif_nodes_values_device(raw)
+                ncount = raw["ncount"]
+                extra["values"] = values_t
+                extra["hidx"] = raw["hidx"]
+                extra["hw"] = raw["hw"]
+            else:
+                packed, values, max_depth = _eif_nodes_values(forest)
+                aos = torch.from_numpy(packed).to(device)
+                ncount = torch.from_numpy(
+                    np.ascontiguousarray(forest.node_count, dtype=np.int32)
+                ).to(device)
+                extra["values"] = torch.from_numpy(values).to(device)
+                extra["hidx"] = torch.from_numpy(
+                    np.ascontiguousarray(forest.hyper_idx)).to(device)
+                extra["hw"] = torch.from_numpy(
+                    np.ascontiguousarray(forest.hyper_w)).to(device)
+            extra["height"] = max_depth
+        elif isinstance(v4_key, tuple) and v4_key[0] == "eif_dense":
             D = v4_key[1]
             if raw is not None:
                 aos, values_t, hw_t, max_depth = _eif_dense_packed_device(
@@ -499,16 +531,48 @@ def score_forest(model, X: torch.Tensor, finalize: bool = True) -> torch.Tensor:
     )
 
 
+def _eif_uniform_nnz(forest) -> bool:
+    """True when every internal node carries exactly forest.nnz coordinates
+    (always the case for forests we build; foreign ragged files route to the
+    general kernel so trailing zero-terms cannot perturb the strict-order
+    dot)."""
+    u = getattr(forest, "_uniform_nnz", None)
+    if u is None:
+        f = forest.feature
+        internal = f >= 0
+        u = bool((f[internal] == forest.nnz).all()) if internal.any() else True
+        forest._uniform_nnz = u
+    return u
+
+
 def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torch.Tensor:
     ext = load_extension()
     forest = model.forest
     c = float(avg_path_length(forest.num_samples))
     d = int(X.shape[1])
     nnz = forest.nnz
-    # densified-dense walk wins over the strict-order sparse kernel once
-    # hyperplanes carry more than ~6 coordinates (measured crossover,
-    # profiles/r01_bench_and_kernels.md); the sparse kernel stays for
-    # low extension levels and wide feature spaces.
+    # routing (measured crossovers, profiles/r01_bench_and_kernels.md):
+    # nnz <= 5 with uniform hyperplane widths -> fixed-trip sparse v2;
+    # wider hyperplanes with d <= 32 -> densified dense v2 walk;
+    # everything else -> the general strict-order kernel.
+    if nnz <= 5 and _eif_uniform_nnz(forest):
+        elem = 2 if X.dtype == torch.bfloat16 else 4
+        dpad = d
+        if elem == 2:
+            while dpad % 4 != 2:
+                dpad += 1
+        else:
+            while dpad % 2 != 1:
+                dpad += 1
+        mn = forest.feature.shape[1]
+        lds = mn * (12 + nnz * 8) + 2 * 256 * dpad * elem
+        if lds <= 150 * 1024:
+            aos, ncount, extra = _device_forest(
+                model, X.device, v4_key="eif_sparse")
+            return ext.score_extended_sparse_v2(
+                X.contiguous(), aos, extra["values"], extra["hidx"],
+                extra["hw"], ncount, extra["height"], c, finalize,
+            )
     if d <= 32 and (nnz == d or nnz >= 6):
         D = 8 if d <= 8 else (16 if d <= 16 else 32)
         aos, ncount, extra = _device_forest(

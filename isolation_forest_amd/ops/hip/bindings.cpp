@@ -58,6 +58,16 @@ void launch_score_extended_forest(bool bf16, bool rows_lds, bool hyper_lds,
                                   int finalize, size_t lds, int blocks,
                                   hipStream_t stream);
 
+void launch_score_extended_sparse_v2(bool bf16, int nnz, const void* X,
+                                     const void* nodes, const float* values,
+                                     const int32_t* hidx, const float* hw,
+                                     const int32_t* ncount, float* out,
+                                     int64_t N, int32_t d, int32_t dpad,
+                                     int32_t T, int32_t max_nodes,
+                                     int32_t height_limit, float fT,
+                                     float c_norm, int finalize, size_t lds,
+                                     int blocks, hipStream_t stream);
+
 void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
                                     const void* nodes, const float* values,
                                     const float* hw, const int32_t* ncount,
@@ -368,6 +378,54 @@ torch::Tensor score_extended_dense_v2(torch::Tensor X,
   return out;
 }
 
+torch::Tensor score_extended_sparse_v2(torch::Tensor X,
+                                       torch::Tensor nodes_packed,
+                                       torch::Tensor values,
+                                       torch::Tensor hidx, torch::Tensor hw,
+                                       torch::Tensor ncount,
+                                       int64_t height_limit, double c_norm,
+                                       bool finalize) {
+  CHECK_CUDA(X);
+  CHECK_CONTIG(X);
+  CHECK_CUDA(nodes_packed);
+  CHECK_CONTIG(nodes_packed);
+  CHECK_CUDA(values);
+  CHECK_CONTIG(values);
+  CHECK_CUDA(hidx);
+  CHECK_CONTIG(hidx);
+  CHECK_CUDA(hw);
+  CHECK_CONTIG(hw);
+  CHECK_CUDA(ncount);
+  check_x(X);
+  int64_t N = X.size(0), d = X.size(1);
+  int64_t T = nodes_packed.size(0), max_nodes = nodes_packed.size(1);
+  int64_t nnz = hidx.size(2);
+  TORCH_CHECK(nnz >= 1 && nnz <= 5, "sparse v2 supports nnz <= 5");
+  auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
+  if (N == 0) return out;
+
+  const bool bf16 = is_bf16(X);
+  const size_t elem = bf16 ? 2 : 4;
+  int64_t dpad = d;
+  if (bf16) {
+    while (dpad % 4 != 2) ++dpad;
+  } else {
+    while (dpad % 2 != 1) ++dpad;
+  }
+  size_t lds = (size_t)max_nodes * (12 + nnz * 8)
+               + (size_t)2 * 256 * dpad * elem;
+  TORCH_CHECK(lds <= 150 * 1024, "sparse v2 LDS overflow; use general path");
+  int blocks = (int)std::min<int64_t>((N + 511) / 512, 8192);
+  ifa::launch_score_extended_sparse_v2(
+      bf16, (int)nnz, X.data_ptr(), nodes_packed.data_ptr<int32_t>(),
+      values.data_ptr<float>(), hidx.data_ptr<int32_t>(),
+      hw.data_ptr<float>(), ncount.data_ptr<int32_t>(),
+      out.data_ptr<float>(), N, (int32_t)d, (int32_t)dpad, (int32_t)T,
+      (int32_t)max_nodes, (int32_t)height_limit, (float)T, (float)c_norm,
+      finalize ? 1 : 0, lds, blocks, current_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bag_gather", &bag_gather, "gather per-tree bags (K9/K10)");
   m.def("build_forest", &build_forest, "build standard iTrees (K1/K2/K11)");
@@ -376,6 +434,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("score_forest", &score_forest, "batched path-length scoring (K6)");
   m.def("score_extended_forest", &score_extended_forest,
         "batched EIF scoring (K7)");
+  m.def("score_extended_sparse_v2", &score_extended_sparse_v2,
+        "sparse EIF scoring, fixed-trip batched walk (K7 small-nnz path)");
   m.def("score_extended_dense_v2", &score_extended_dense_v2,
         "dense EIF scoring, rows-in-registers (K7 fast path)");
   m.attr("WAVE") = 64;

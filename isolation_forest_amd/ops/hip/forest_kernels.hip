@@ -921,6 +921,128 @@ __global__ void __launch_bounds__(EIFD_THREADS, 4) score_extended_dense_v2(
 }
 
 // ---------------------------------------------------------------------------
+// extended scoring, sparse fast path v2 (nnz <= 5, templated):
+// same node/value packing as the dense v2 kernel (self-looping leaves with
+// offset -inf and zero weights => fixed-trip walks), rows staged native in
+// LDS, hyperplanes (idx+w) staged in LDS, RPT=2 chains per thread with
+// batch-phased reads. The dot keeps the ORACLE's strict j-order f32
+// accumulation (bitwise vs cpu_engine.path_lengths_extended).
+// ---------------------------------------------------------------------------
+
+template <typename KT, int NNZ, int RPT>
+__global__ void __launch_bounds__(256) score_extended_sparse_v2(
+    const KT* __restrict__ X,           // raw bits [N][d]
+    const int2* __restrict__ nodes,     // [T][max_nodes] {w0, offset/-inf}
+    const float* __restrict__ values,   // [T][max_nodes] leaf value+depth
+    const int32_t* __restrict__ hidx_g, // [T][max_nodes][NNZ]
+    const float* __restrict__ hw_g,     // [T][max_nodes][NNZ]
+    const int32_t* __restrict__ ncnt, float* __restrict__ out, int64_t N,
+    int32_t d, int32_t dpad, int32_t T, int32_t max_nodes,
+    int32_t height_limit, float fT, float c_norm, int32_t finalize) {
+  const int tid = threadIdx.x;
+  const int rows_per_iter = RPT * 256;
+
+  int2* tlds = (int2*)smem;                   // [max_nodes]
+  float* vlds = (float*)(tlds + max_nodes);   // [max_nodes]
+  int32_t* ilds = (int32_t*)(vlds + max_nodes);  // [max_nodes][NNZ]
+  float* wlds = (float*)(ilds + max_nodes * NNZ);  // [max_nodes][NNZ]
+  KT* rows = (KT*)(wlds + max_nodes * NNZ);   // [rows_per_iter][dpad]
+
+  for (int64_t block_row0 = (int64_t)blockIdx.x * rows_per_iter; block_row0 < N;
+       block_row0 += (int64_t)gridDim.x * rows_per_iter) {
+    const int rows_here = (int)min((int64_t)rows_per_iter, N - block_row0);
+    __syncthreads();  // previous iteration's readers done
+    {
+      const int64_t total = (int64_t)rows_here * d;
+      for (int64_t g = tid; g < total; g += 256) {
+        const int r = (int)(g / (uint32_t)d), c = (int)(g % (uint32_t)d);
+        rows[r * dpad + c] = X[(block_row0 + r) * d + c];
+      }
+    }
+    __syncthreads();
+    const KT* rb[RPT];
+#pragma unroll
+    for (int r = 0; r < RPT; ++r) rb[r] = rows + (tid + r * 256) * dpad;
+
+    float psum[RPT];
+#pragma unroll
+    for (int r = 0; r < RPT; ++r) psum[r] = 0.f;
+
+    for (int t = 0; t < T; ++t) {
+      __syncthreads();
+      const int nc = ncnt[t];
+      {
+        const int2* ss = nodes + (int64_t)t * max_nodes;
+        const float* vs = values + (int64_t)t * max_nodes;
+        for (int i = tid; i < nc; i += 256) {
+          tlds[i] = ss[i];
+          vlds[i] = vs[i];
+        }
+        const int32_t* is = hidx_g + (int64_t)t * max_nodes * NNZ;
+        const float* ws = hw_g + (int64_t)t * max_nodes * NNZ;
+        for (int g = tid; g < nc * NNZ; g += 256) {
+          ilds[g] = is[g];
+          wlds[g] = ws[g];
+        }
+      }
+      __syncthreads();
+
+      int cur[RPT];
+#pragma unroll
+      for (int r = 0; r < RPT; ++r) cur[r] = 0;
+
+      for (int it = 0; it < height_limit; ++it) {
+        int2 nd[RPT];
+#pragma unroll
+        for (int r = 0; r < RPT; ++r) nd[r] = tlds[cur[r]];
+        int32_t ci[RPT][NNZ];
+#pragma unroll
+        for (int r = 0; r < RPT; ++r)
+#pragma unroll
+          for (int j = 0; j < NNZ; ++j) ci[r][j] = ilds[cur[r] * NNZ + j];
+        float cw[RPT][NNZ];
+#pragma unroll
+        for (int r = 0; r < RPT; ++r)
+#pragma unroll
+          for (int j = 0; j < NNZ; ++j) cw[r][j] = wlds[cur[r] * NNZ + j];
+        float xv[RPT][NNZ];
+#pragma unroll
+        for (int r = 0; r < RPT; ++r)
+#pragma unroll
+          for (int j = 0; j < NNZ; ++j)
+            xv[r][j] = load_row_f32<KT>(rb[r], ci[r][j]);
+#pragma unroll
+        for (int r = 0; r < RPT; ++r) {
+          float dot = 0.f;
+#pragma unroll
+          for (int j = 0; j < NNZ; ++j)  // oracle j-order
+            dot = __fadd_rn(dot, __fmul_rn(cw[r][j], xv[r][j]));
+          const int right = pn_right(nd[r].x);
+          cur[r] = (dot < __int_as_float(nd[r].y)) ? cur[r] + 1 : right;
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < RPT; ++r)
+        psum[r] = __fadd_rn(psum[r], vlds[cur[r]]);
+    }
+
+#pragma unroll
+    for (int r = 0; r < RPT; ++r) {
+      const int64_t my_row = block_row0 + tid + r * 256;
+      if (my_row < N) {
+        if (finalize) {
+          const float mean32 = (float)((double)psum[r] / (double)fT);
+          const double ratio = (double)mean32 / (double)c_norm;
+          out[my_row] = (float)exp2(-ratio);
+        } else {
+          out[my_row] = psum[r];
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // extended scoring, sparse general path (nnz < d): exact oracle order.
 // ---------------------------------------------------------------------------
 
@@ -1162,6 +1284,39 @@ void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
     else LSD2(uint32_t, 32, 1);
   }
 #undef LSD2
+}
+
+void launch_score_extended_sparse_v2(bool bf16, int nnz, const void* X,
+                                     const void* nodes, const float* values,
+                                     const int32_t* hidx, const float* hw,
+                                     const int32_t* ncount, float* out,
+                                     int64_t N, int32_t d, int32_t dpad,
+                                     int32_t T, int32_t max_nodes,
+                                     int32_t height_limit, float fT,
+                                     float c_norm, int finalize, size_t lds,
+                                     int blocks, hipStream_t stream) {
+#define LSS2(KT, NZ)                                                          \
+  do {                                                                        \
+    raise_lds((const void*)score_extended_sparse_v2<KT, NZ, 2>, lds);         \
+    hipLaunchKernelGGL((score_extended_sparse_v2<KT, NZ, 2>), dim3(blocks),   \
+                       dim3(256), lds, stream, (const KT*)X,                  \
+                       (const int2*)nodes, values, hidx, hw, ncount, out, N,  \
+                       d, dpad, T, max_nodes, height_limit, fT, c_norm,       \
+                       finalize);                                             \
+  } while (0)
+#define LSS2_ALL(KT)                                                          \
+  do {                                                                        \
+    switch (nnz) {                                                            \
+      case 1: LSS2(KT, 1); break;                                             \
+      case 2: LSS2(KT, 2); break;                                             \
+      case 3: LSS2(KT, 3); break;                                             \
+      case 4: LSS2(KT, 4); break;                                             \
+      default: LSS2(KT, 5); break;                                            \
+    }                                                                         \
+  } while (0)
+  if (bf16) LSS2_ALL(uint16_t); else LSS2_ALL(uint32_t);
+#undef LSS2_ALL
+#undef LSS2
 }
 
 void launch_score_extended_forest(bool bf16, bool rows_lds, bool hyper_lds,

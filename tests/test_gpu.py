@@ -293,3 +293,34 @@ class TestDevicePacking:
         s_host_pack = model.score(X).cpu().numpy()
         np.testing.assert_array_equal(
             s_dev_pack.view(np.int32), s_host_pack.view(np.int32))
+
+
+class TestSparseV2:
+    def test_sparse_path_sums_bitwise(self, dev):
+        """nnz <= 5 routes through the fixed-trip sparse v2 kernel, which
+        keeps the oracle's strict j-order dot: bitwise path sums."""
+        from isolation_forest_amd.ops import gpu_engine
+
+        X = make_data(3000, 6, seed=31)
+        bag = cpu_engine.sample_bags(3000, 24, 128, seed=8, bootstrap=False)
+        fs = cpu_engine.feature_subsets(6, 6, 24, seed=8)
+        forest = cpu_engine.build_extended_forest(X, bag, fs, 8, 128, 6, 6, 3)
+        cpu_ps = cpu_engine.path_lengths_extended(forest, X)
+        model = ExtendedIsolationForest(numEstimators=24).fit(X[:600])
+        model.forest = forest
+        model._gpu_forest_cache = {}
+        gpu_ps = gpu_engine.score_extended_forest(
+            model, torch.from_numpy(X).to(dev), finalize=False
+        )
+        np.testing.assert_array_equal(
+            gpu_ps.cpu().numpy().view(np.int32), cpu_ps.view(np.int32)
+        )
+
+    def test_sparse_end_to_end_gpu_fit(self, dev):
+        X = torch.from_numpy(make_data(20000, 10, seed=32)).to(dev)
+        model = ExtendedIsolationForest(
+            numEstimators=50, extensionLevel=2, randomSeed=4).fit(X)
+        scores = model.score(X)
+        cpu_scores = model.score(X.float().cpu()).numpy()
+        np.testing.assert_array_equal(
+            scores.cpu().numpy().view(np.int32), cpu_scores.view(np.int32))
