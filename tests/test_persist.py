@@ -270,3 +270,44 @@ class TestFastCodec:
         _, records = avro_io.read_container(f)
         assert len(records) == int(model.forest.node_count.sum())
         assert records[0]["nodeData"]["id"] == 0
+
+
+class TestDegenerateModels:
+    """Reference: IsolationForestModelWriteReadTest.scala:251-344 — an
+    empty-forest model round-trips but transform throws; numSamples=1
+    transform throws."""
+
+    def test_empty_forest_roundtrip_then_transform_throws(self, tmp_path):
+        from isolation_forest_amd import IsolationForestModel
+        from isolation_forest_amd.core.forest import empty_forest
+        from isolation_forest_amd.utils.params import Params
+
+        forest = empty_forest(0, 1, num_samples=256, num_features=4,
+                              total_num_features=4)
+        model = IsolationForestModel(uid="empty-test", forest=forest,
+                                     params=Params())
+        p = str(tmp_path / "empty")
+        model.save(p)
+        loaded = IsolationForestModel.load(p)
+        assert loaded.forest.num_trees == 0
+        X = torch.zeros((5, 4))
+        with pytest.raises(ValueError, match="no trees"):
+            loaded.transform(X)
+
+    def test_num_samples_one_transform_throws(self, tmp_path):
+        from isolation_forest_amd import IsolationForestModel
+        from isolation_forest_amd.core.forest import empty_forest
+        from isolation_forest_amd.utils.params import Params
+
+        forest = empty_forest(1, 1, num_samples=1, num_features=4,
+                              total_num_features=4)
+        forest.node_count[0] = 1
+        forest.feature[0, 0] = -1  # single leaf
+        forest.num_instances[0, 0] = 1
+        model = IsolationForestModel(uid="one-sample", forest=forest,
+                                     params=Params())
+        p = str(tmp_path / "one")
+        model.save(p)
+        loaded = IsolationForestModel.load(p)
+        with pytest.raises(ValueError, match=">= 2 required"):
+            loaded.transform(torch.zeros((3, 4)))
