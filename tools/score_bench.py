@@ -37,9 +37,15 @@ def main():
     dev = "cuda:0"
     g = torch.Generator(device=dev)
     g.manual_seed(7)
-    X = torch.randn((args.rows, args.features), device=dev, generator=g)
-    if args.dtype == "bf16":
-        X = X.to(torch.bfloat16)
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    # chunked generation keeps peak memory ~= the final matrix (1B x 32
+    # bf16 = 64 GB fits one MI355X with room to spare)
+    X = torch.empty((args.rows, args.features), device=dev, dtype=dtype)
+    step = max(1, min(args.rows, 64_000_000))
+    for lo in range(0, args.rows, step):
+        hi = min(args.rows, lo + step)
+        X[lo:hi] = torch.randn((hi - lo, args.features), device=dev,
+                               generator=g).to(dtype)
 
     cls = ExtendedIsolationForest if args.extended else IsolationForest
     kw = {}
