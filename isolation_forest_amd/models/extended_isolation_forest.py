@@ -8,16 +8,14 @@ estimator (no leakage across fits — ExtendedIsolationForest.scala:102).
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from ..core import cpu_engine
 from ..core.forest import ExtendedForest
-from ..utils.params import ExtendedParams, resolve_params
+from ..utils.params import ExtendedParams
 from . import base
-from .base import ModelBase, new_uid
-from .isolation_forest import IsolationForest, _shard_range
+from .base import ModelBase
+from .isolation_forest import IsolationForest
 
 
 class ExtendedIsolationForest(IsolationForest):

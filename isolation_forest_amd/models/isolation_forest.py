@@ -12,7 +12,6 @@ from __future__ import annotations
 import logging
 from typing import Optional
 
-import numpy as np
 import torch
 
 from ..core import cpu_engine

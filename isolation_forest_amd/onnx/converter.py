@@ -22,7 +22,7 @@ from __future__ import annotations
 
 import json
 import math
-from typing import Dict, List
+from typing import Dict
 
 from ..persist import avro_io
 from . import model_proto as mp

@@ -14,7 +14,7 @@ value promoted), aggregate AVERAGE = mean of per-tree leaf weights.
 
 from __future__ import annotations
 
-from typing import Dict, List
+from typing import Dict
 
 import numpy as np
 
