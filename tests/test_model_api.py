@@ -201,3 +201,44 @@ class TestMammographyQuality:
         ).fit(X)
         score = model.transform(X)["outlierScore"].numpy()
         assert auroc(y, score) == pytest.approx(0.86, abs=0.02)
+
+
+class TestExactQuantileStress:
+    """compute_threshold exact mode must return the true order statistic
+    even on tie-heavy and tightly-clustered score distributions (the
+    histogram-refinement path with comparator-anchored ranks)."""
+
+    @pytest.mark.parametrize("case", ["uniform", "ties", "clustered", "tiny_range"])
+    def test_exact_matches_sorted(self, case):
+        from isolation_forest_amd.core import threshold as th
+
+        rs = np.random.RandomState(hash(case) % 2**31)
+        if case == "uniform":
+            s = rs.uniform(0, 1, 300_000).astype(np.float32)
+        elif case == "ties":
+            s = rs.choice(
+                np.linspace(0.3, 0.7, 50).astype(np.float32), 300_000)
+        elif case == "clustered":
+            s = np.concatenate([
+                rs.normal(0.5, 1e-6, 290_000),
+                rs.uniform(0, 1, 10_000)]).astype(np.float32)
+        else:
+            s = (0.5 + rs.uniform(-1e-7, 1e-7, 300_000)).astype(np.float32)
+        old = th._EXACT_KTH_CUTOFF
+        th._EXACT_KTH_CUTOFF = 1000  # force the histogram-refinement path
+        try:
+            for contamination in [0.02, 0.25, 0.001]:
+                t = torch.from_numpy(s)
+                got = th.compute_threshold(t, contamination, 0.0)
+                k = max(1, min(len(s),
+                               int(np.ceil((1 - contamination) * len(s)))))
+                expect = float(np.sort(s)[k - 1])
+                assert got == pytest.approx(expect, abs=0), (case, contamination)
+        finally:
+            th._EXACT_KTH_CUTOFF = old
+        for contamination in [0.02]:
+            t = torch.from_numpy(s)
+            got = th.compute_threshold(t, contamination, 0.0)
+            k = max(1, min(len(s), int(np.ceil((1 - contamination) * len(s)))))
+            expect = float(np.sort(s)[k - 1])
+            assert got == pytest.approx(expect, abs=0), (case, contamination)
