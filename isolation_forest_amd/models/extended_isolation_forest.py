@@ -49,7 +49,7 @@ class ExtendedIsolationForest(IsolationForest):
     def _gather_forest(self, forest_local, comm, rp, total_features):
         if comm is None:
             return forest_local
-        arrays = comm.all_gather_forest_arrays(
+        named = dict(
             feature=forest_local.feature,
             value=forest_local.value,
             right=forest_local.right,
@@ -59,13 +59,21 @@ class ExtendedIsolationForest(IsolationForest):
             hyper_w=forest_local.hyper_w,
             offset64=forest_local.offset64,
         )
-        return ExtendedForest(
+        depth_np = getattr(forest_local, "depth_np", None)
+        if depth_np is not None:
+            named["depth_np"] = depth_np
+        arrays = comm.all_gather_forest_arrays(**named)
+        depth_np = arrays.pop("depth_np", None)
+        forest = ExtendedForest(
             num_samples=forest_local.num_samples,
             num_features=forest_local.num_features,
             total_num_features=forest_local.total_num_features,
             extension_level=forest_local.extension_level,
             **arrays,
         )
+        if depth_np is not None:
+            forest.depth_np = depth_np
+        return forest
 
 
 class ExtendedIsolationForestModel(ModelBase):

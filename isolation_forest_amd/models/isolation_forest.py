@@ -143,7 +143,7 @@ class IsolationForest:
     def _gather_forest(self, forest_local, comm, rp, total_features) -> Forest:
         if comm is None:
             return forest_local
-        arrays = comm.all_gather_forest_arrays(
+        named = dict(
             feature=forest_local.feature,
             value=forest_local.value,
             right=forest_local.right,
@@ -151,12 +151,20 @@ class IsolationForest:
             node_count=forest_local.node_count,
             value64=forest_local.value64,
         )
-        return Forest(
+        depth_np = getattr(forest_local, "depth_np", None)
+        if depth_np is not None:  # GPU builds: keep device packing possible
+            named["depth_np"] = depth_np
+        arrays = comm.all_gather_forest_arrays(**named)
+        depth_np = arrays.pop("depth_np", None)
+        forest = Forest(
             num_samples=forest_local.num_samples,
             num_features=forest_local.num_features,
             total_num_features=forest_local.total_num_features,
             **arrays,
         )
+        if depth_np is not None:
+            forest.depth_np = depth_np
+        return forest
 
     # spark.ml parity helper
     def transformSchema(self, columns):

@@ -67,11 +67,14 @@ def build_forest(X: torch.Tensor, bag_idx: np.ndarray, feat_sub: np.ndarray,
         total_num_features=rp.total_features,
         value64=np.where(f_np >= 0, v_np, 0.0).astype(np.float64),
     )
-    # raw device tensors for scoring-pack reuse (skips the host round trip)
+    # raw device tensors for scoring-pack reuse (skips the host round trip);
+    # depth_np additionally travels through the forest all-gather so
+    # multi-rank fits keep the device packing path
     forest._device_raw = {
         "device": str(device), "feat": feat, "value": value, "right": right,
         "ncount": ncount, "depth": depth,
     }
+    forest.depth_np = depth.cpu().numpy()
     return forest
 
 
@@ -124,6 +127,7 @@ def build_extended_forest(X: torch.Tensor, bag_idx: np.ndarray,
         "device": str(device), "feat": feat, "value": value, "right": right,
         "ncount": ncount, "depth": depth, "hidx": hidx, "hw": hw,
     }
+    fr.depth_np = depth.cpu().numpy()
     return fr
 
 
@@ -454,6 +458,29 @@ def _device_forest(model, device, v4_key=None):
         raw = getattr(forest, "_device_raw", None)
         if raw is not None and raw["device"] != str(device):
             raw = None
+        if raw is None and device.type == "cuda" and \
+                getattr(forest, "depth_np", None) is not None:
+            # e.g. an all-gathered multi-rank forest: rebuild the raw view
+            # from host arrays (one upload) so packing stays on device
+            raw = {
+                "device": str(device),
+                "feat": torch.from_numpy(
+                    np.ascontiguousarray(forest.feature)).to(device),
+                "value": torch.from_numpy(
+                    np.ascontiguousarray(forest.value)).to(device),
+                "right": torch.from_numpy(
+                    np.ascontiguousarray(forest.right)).to(device),
+                "ncount": torch.from_numpy(np.ascontiguousarray(
+                    forest.node_count, dtype=np.int32)).to(device),
+                "depth": torch.from_numpy(np.ascontiguousarray(
+                    forest.depth_np, dtype=np.int32)).to(device),
+            }
+            if isinstance(forest, ExtendedForest):
+                raw["hidx"] = torch.from_numpy(
+                    np.ascontiguousarray(forest.hyper_idx)).to(device)
+                raw["hw"] = torch.from_numpy(
+                    np.ascontiguousarray(forest.hyper_w)).to(device)
+            forest._device_raw = raw
         if v4_key == "eif_sparse":
             if raw is not None:
                 aos, values_t, _, max_depth = _eif_nodes_values_device(raw)

@@ -287,8 +287,9 @@ class TestDevicePacking:
         X = torch.from_numpy(make_data(30000, 10, seed=23)).to(dev)
         model = IsolationForest(numEstimators=64, randomSeed=9).fit(X)
         s_dev_pack = model.score(X).cpu().numpy()
-        # force the host packing path
+        # force the host packing path (drop raw AND the rebuildable depths)
         del model.forest._device_raw
+        model.forest.depth_np = None
         model._gpu_forest_cache = {}
         s_host_pack = model.score(X).cpu().numpy()
         np.testing.assert_array_equal(
