@@ -242,3 +242,49 @@ class TestExactQuantileStress:
             k = max(1, min(len(s), int(np.ceil((1 - contamination) * len(s)))))
             expect = float(np.sort(s)[k - 1])
             assert got == pytest.approx(expect, abs=0), (case, contamination)
+
+
+class TestSklearnAdapter:
+    def test_contract(self, gaussian_data):
+        from sklearn.metrics import roc_auc_score
+        from isolation_forest_amd.sklearn_api import IsolationForestSKL
+
+        X, y = gaussian_data
+        clf = IsolationForestSKL(n_estimators=60, contamination=0.05,
+                                 random_state=3)
+        labels = clf.fit_predict(X)
+        assert set(np.unique(labels)) <= {-1, 1}
+        s = clf.score_samples(X)
+        assert s.shape == (len(X),)
+        # sklearn convention: LOWER score_samples = more anomalous
+        assert roc_auc_score(y, -s) > 0.9
+        d = clf.decision_function(X)
+        np.testing.assert_array_equal(labels, np.where(d < 0, -1, 1))
+        # flagged fraction tracks contamination
+        frac = float((labels == -1).mean())
+        assert frac == pytest.approx(0.05, abs=0.02)
+
+    def test_auto_contamination_and_params(self, gaussian_data):
+        from isolation_forest_amd.sklearn_api import IsolationForestSKL
+
+        X, y = gaussian_data
+        clf = IsolationForestSKL(random_state=5)
+        assert clf.get_params()["contamination"] == "auto"
+        clf.set_params(n_estimators=40)
+        clf.fit(X)
+        assert clf.offset_ == -0.5
+        labels = clf.predict(X)
+        assert (labels == -1).any() or (labels == 1).all()
+
+    def test_sklearn_pipeline_compatible(self, gaussian_data):
+        from sklearn.pipeline import Pipeline
+        from sklearn.preprocessing import StandardScaler
+        from isolation_forest_amd.sklearn_api import IsolationForestSKL
+
+        X, y = gaussian_data
+        pipe = Pipeline([
+            ("scale", StandardScaler()),
+            ("iforest", IsolationForestSKL(n_estimators=40, random_state=2)),
+        ])
+        labels = pipe.fit_predict(X)
+        assert labels.shape == (len(X),)
