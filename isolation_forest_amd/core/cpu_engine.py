@@ -290,21 +290,25 @@ def _draw_hyperplane(
     ExtendedIsolationTree.scala:183-184)."""
     k = len(features)
     avail = features.copy()
+    jarr = np.arange(nnz, dtype=np.uint32)
+    # batched Philox draws — identical counters (and therefore bitwise-
+    # identical values) to per-j scalar calls, ~5x faster per node
+    fy = rng.randint_below(
+        seed, rng.P_EIF_COORD, np.uint32(tree_id), np.uint32(node_id),
+        (k - jarr.astype(np.int64)), attempt=jarr,
+    )
     for j in range(nnz):
-        t = j + int(
-            rng.randint_below(
-                seed, rng.P_EIF_COORD, np.uint32(tree_id), np.uint32(node_id),
-                k - j, attempt=np.uint32(j),
-            )
-        )
+        t = j + int(fy[j])
         avail[j], avail[t] = avail[t], avail[j]
     coords = np.array(sorted(int(c) for c in avail[:nnz]), dtype=np.int32)
 
     # weights drawn AFTER sorting, keyed by slot j over sorted coords
-    w = np.empty(nnz, dtype=np.float32)
     base = node_id * 4096  # nnz-slot stride; nnz <= 4096 by construction
-    for j in range(nnz):
-        w[j] = np.float32(_gaussian(seed, tree_id, base + j))
+    u1, u2 = rng.uniform2(
+        seed, rng.P_EIF_NORMAL, np.uint32(tree_id),
+        np.uint32(base) + jarr,
+    )
+    w = det_math.det_gaussian(u1, u2).astype(np.float32)
     # L2 normalize in float32, sequential accumulation
     acc = np.float32(0.0)
     for j in range(nnz):
@@ -315,19 +319,17 @@ def _draw_hyperplane(
     w = (w / norm).astype(np.float32)
 
     # intercepts per sorted coordinate; offset in float64
+    sub = bag[np.ix_(seg, coords)]
+    mn = sub.min(axis=0).astype(np.float32).astype(np.float64)
+    mx = sub.max(axis=0).astype(np.float32).astype(np.float64)
+    u = rng.uniform(
+        seed, rng.P_EIF_INTERCEPT, np.uint32(tree_id),
+        np.uint32(base) + jarr,
+    ).astype(np.float64)
+    intercepts = mn + u * (mx - mn)
     off64 = 0.0
     for j in range(nnz):
-        col = bag[seg, coords[j]]
-        mn = np.float32(col.min())
-        mx = np.float32(col.max())
-        u = float(
-            rng.uniform(
-                seed, rng.P_EIF_INTERCEPT, np.uint32(tree_id),
-                np.uint32(base + j),
-            )
-        )
-        intercept = float(mn) + u * (float(mx) - float(mn))
-        off64 += float(w[j]) * intercept
+        off64 += float(w[j]) * float(intercepts[j])
     return coords, w, off64, np.float32(off64)
 
 
