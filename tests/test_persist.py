@@ -311,3 +311,58 @@ class TestDegenerateModels:
         loaded = IsolationForestModel.load(p)
         with pytest.raises(ValueError, match=">= 2 required"):
             loaded.transform(torch.zeros((3, 4)))
+
+
+class TestRaggedExtendedRoundtrip:
+    """Foreign extended models may carry hyperplanes of varying widths;
+    the fast codec must round-trip them exactly (masked columns)."""
+
+    def test_mixed_nnz_roundtrip(self, tmp_path):
+        from isolation_forest_amd import ExtendedIsolationForestModel
+        from isolation_forest_amd.core.forest import empty_extended_forest
+        from isolation_forest_amd.utils.params import ExtendedParams
+
+        rs = np.random.RandomState(4)
+        fr = empty_extended_forest(2, 7, 3, num_samples=64, num_features=5,
+                                   total_num_features=5, extension_level=2)
+        for t in range(2):
+            # root(3 coords) -> [leaf, internal(2 coords) -> [leaf, leaf]]
+            fr.node_count[t] = 5
+            fr.feature[t, 0] = 3
+            fr.hyper_idx[t, 0, :3] = [0, 2, 4]
+            fr.hyper_w[t, 0, :3] = rs.normal(size=3).astype(np.float32)
+            fr.offset64[t, 0] = 0.25 + t
+            fr.right[t, 0] = 2
+            fr.num_instances[t, 0] = -1
+            fr.feature[t, 1] = -1
+            fr.num_instances[t, 1] = 30
+            fr.feature[t, 2] = 2
+            fr.hyper_idx[t, 2, :2] = [1, 3]
+            fr.hyper_w[t, 2, :2] = rs.normal(size=2).astype(np.float32)
+            fr.offset64[t, 2] = -0.5
+            fr.right[t, 2] = 4
+            fr.num_instances[t, 2] = -1
+            for leaf_i, cnt in [(3, 20), (4, 14)]:
+                fr.feature[t, leaf_i] = -1
+                fr.num_instances[t, leaf_i] = cnt
+        model = ExtendedIsolationForestModel(
+            uid="ragged", forest=fr, params=ExtendedParams())
+        p = str(tmp_path / "ragged")
+        model.save(p)
+        loaded = ExtendedIsolationForestModel.load(p)
+        lf = loaded.forest
+        for t in range(2):
+            assert lf.feature[t, 0] == 3 and lf.feature[t, 2] == 2
+            np.testing.assert_array_equal(lf.hyper_idx[t, 0, :3],
+                                          fr.hyper_idx[t, 0, :3])
+            np.testing.assert_array_equal(
+                lf.hyper_w[t, 2, :2].view(np.int32),
+                fr.hyper_w[t, 2, :2].view(np.int32))
+            assert lf.offset64[t, 0] == fr.offset64[t, 0]
+        # generic reader agrees with the fast-written bytes
+        from isolation_forest_amd.persist import avro_io
+        f = glob.glob(os.path.join(p, "data", "*.avro"))[0]
+        _, records = avro_io.read_container(f)
+        assert len(records) == 10
+        r0 = records[0]["extendedNodeData"]
+        assert r0["indices"] == [0, 2, 4]
