@@ -325,3 +325,47 @@ class TestSparseV2:
         cpu_scores = model.score(X.float().cpu()).numpy()
         np.testing.assert_array_equal(
             scores.cpu().numpy().view(np.int32), cpu_scores.view(np.int32))
+
+
+class TestWideFeatureFallbacks:
+    """d large enough that row tiles exceed LDS exercises the global-X
+    walk (key transform on the fly) and the general EIF kernel."""
+
+    def test_standard_wide_d_bitwise(self, dev):
+        from isolation_forest_amd.ops import gpu_engine
+
+        d = 700  # bf16 row tile > 150 KB => ROWS_LDS=false path
+        X = make_data(3000, d, seed=41)
+        bag = cpu_engine.sample_bags(3000, 8, 128, seed=6, bootstrap=False)
+        fs = cpu_engine.feature_subsets(d, 32, 8, seed=6)
+        forest = cpu_engine.build_forest(X, bag, fs, 6, 128, 32, d)
+        cpu_ps = cpu_engine.path_lengths(forest, X)
+        model = IsolationForest(numEstimators=8).fit(X[:500])
+        model.forest = forest
+        model._gpu_forest_cache = {}
+        for dt in (torch.float32, torch.bfloat16):
+            Xt = torch.from_numpy(X).to(dev).to(dt)
+            ref = cpu_ps if dt == torch.float32 else cpu_engine.path_lengths(
+                forest, Xt.float().cpu().numpy())
+            gpu_ps = gpu_engine.score_forest(model, Xt, finalize=False)
+            np.testing.assert_array_equal(
+                gpu_ps.cpu().numpy().view(np.int32), ref.view(np.int32))
+
+    def test_extended_wide_d_small_nnz(self, dev):
+        X = torch.from_numpy(make_data(20000, 40, seed=42)).to(dev)
+        model = ExtendedIsolationForest(
+            numEstimators=20, extensionLevel=2, randomSeed=5).fit(X)
+        s_gpu = model.score(X).cpu().numpy()
+        s_cpu = model.score(X.float().cpu()).numpy()
+        np.testing.assert_array_equal(
+            s_gpu.view(np.int32), s_cpu.view(np.int32))
+
+    def test_extended_wide_d_wide_nnz(self, dev):
+        """d=40 > 32 with nnz=40 routes to the general strict-order
+        kernel (no dense v2): bitwise."""
+        X = torch.from_numpy(make_data(15000, 40, seed=43)).to(dev)
+        model = ExtendedIsolationForest(numEstimators=12, randomSeed=6).fit(X)
+        s_gpu = model.score(X).cpu().numpy()
+        s_cpu = model.score(X.float().cpu()).numpy()
+        diff = np.abs(s_gpu - s_cpu)
+        assert np.quantile(diff, 0.999) < 1e-3
