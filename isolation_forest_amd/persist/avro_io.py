@@ -207,12 +207,50 @@ def _decompress(codec: str, block: bytes) -> bytes:
     raise ValueError(f"unsupported avro codec {codec!r}")
 
 
+def snappy_compress_literal(data: bytes) -> bytes:
+    """Valid (uncompressed) snappy stream: length preamble + one literal
+    element. Any snappy decoder accepts it; used for codec parity with the
+    reference's Spark-written files (which default to snappy)."""
+    out = bytearray()
+    n = len(data)
+    while True:  # LEB128 uncompressed-length preamble
+        b = n & 0x7F
+        n >>= 7
+        out.append(b | (0x80 if n else 0))
+        if not n:
+            break
+    ln = len(data) - 1
+    if len(data) == 0:
+        return bytes(out)
+    if ln < 60:
+        out.append(ln << 2)
+    elif ln < (1 << 8):
+        out.append(60 << 2)
+        out += ln.to_bytes(1, "little")
+    elif ln < (1 << 16):
+        out.append(61 << 2)
+        out += ln.to_bytes(2, "little")
+    elif ln < (1 << 24):
+        out.append(62 << 2)
+        out += ln.to_bytes(3, "little")
+    else:
+        out.append(63 << 2)
+        out += ln.to_bytes(4, "little")
+    out += data
+    return bytes(out)
+
+
 def _compress(codec: str, block: bytes) -> bytes:
     if codec == "null":
         return block
     if codec == "deflate":
         c = zlib.compressobj(level=6, wbits=-15)
         return c.compress(block) + c.flush()
+    if codec == "snappy":
+        # avro snappy codec: payload + 4-byte big-endian CRC32 of the
+        # UNCOMPRESSED bytes (mirror of _decompress)
+        crc = zlib.crc32(block) & 0xFFFFFFFF
+        return snappy_compress_literal(block) + struct.pack(">I", crc)
     raise ValueError(f"unsupported write codec {codec!r}")
 
 

@@ -18,7 +18,7 @@ from isolation_forest_amd import (
     IsolationForest,
     IsolationForestModel,
 )
-from isolation_forest_amd.persist import avro_io
+from isolation_forest_amd.persist import avro_io, model_io
 from tests.conftest import REFERENCE_RESOURCES, auroc
 
 
@@ -366,3 +366,25 @@ class TestRaggedExtendedRoundtrip:
         assert len(records) == 10
         r0 = records[0]["extendedNodeData"]
         assert r0["indices"] == [0, 2, 4]
+
+
+class TestSnappyWrite:
+    def test_snappy_codec_roundtrip(self, trained, tmp_path):
+        model, X = trained
+        p = str(tmp_path / "snappy_model")
+        model_io.save_model(model, p, codec="snappy")
+        f = glob.glob(os.path.join(p, "data", "*.avro"))[0]
+        schema, records = avro_io.read_container(f)
+        assert len(records) == int(model.forest.node_count.sum())
+        loaded = IsolationForestModel.load(p)
+        s1 = model.score(torch.from_numpy(X)).numpy()
+        s2 = loaded.score(torch.from_numpy(X)).numpy()
+        np.testing.assert_array_equal(s1.view(np.int32), s2.view(np.int32))
+
+    def test_snappy_literal_stream(self):
+        from isolation_forest_amd.persist.avro_io import (
+            snappy_compress_literal, snappy_decompress)
+
+        for n in [0, 1, 59, 60, 61, 255, 256, 65535, 65536, 1 << 20]:
+            data = bytes((i * 7 + n) & 0xFF for i in range(n))
+            assert snappy_decompress(snappy_compress_literal(data)) == data
