@@ -75,3 +75,27 @@ class TestBenchContract:
         assert len(lines) == 1, out.stdout
         r = self._check_payload(lines[0], 2)
         assert r["n_gpus"] in (0, 2)  # 0 = CPU plumbing mode
+
+
+class TestTimingSync:
+    def test_sync_env_flag(self, monkeypatch):
+        from isolation_forest_amd.utils.timing import PhaseTimes, phase
+
+        monkeypatch.setenv("IFA_TIMING_SYNC", "1")
+        times = PhaseTimes()
+        with phase(times, "work", device=torch.device("cpu")):
+            pass
+        assert "work" in times
+        # None sink is a no-op
+        with phase(None, "ignored"):
+            pass
+
+    def test_accumulates(self):
+        from isolation_forest_amd.utils.timing import PhaseTimes, phase
+
+        times = PhaseTimes()
+        for _ in range(3):
+            with phase(times, "p"):
+                pass
+        assert times["p"] >= 0.0
+        assert times.total == sum(times.values())
