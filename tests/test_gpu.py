@@ -369,3 +369,64 @@ class TestWideFeatureFallbacks:
         s_cpu = model.score(X.float().cpu()).numpy()
         diff = np.abs(s_gpu - s_cpu)
         assert np.quantile(diff, 0.999) < 1e-3
+
+
+class TestNativeCLI:
+    """tools/native/ifa_score: the torch-free standalone scorer must
+    reproduce the Python engine's scores (same walk decisions; leaf
+    constants may differ by a libm-vs-numpy float32-log ulp)."""
+
+    def test_cli_matches_engine(self, dev, tmp_path):
+        import subprocess
+        import os
+
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        cli = os.path.join(repo, "tools", "native", "ifa_score")
+        if not os.path.exists(cli):
+            subprocess.check_call(
+                ["bash", os.path.join(repo, "tools", "native", "build.sh")])
+        X = make_data(50000, 10, seed=51)
+        model = IsolationForest(
+            numEstimators=100, contamination=0.02, contaminationError=0.01,
+            randomSeed=12).fit(torch.from_numpy(X).to(dev))
+        mdir = str(tmp_path / "m")
+        model.save(mdir)
+        xbin = str(tmp_path / "x.bin")
+        X.astype("<f4").tofile(xbin)
+        sbin = str(tmp_path / "s.bin")
+        lbin = str(tmp_path / "l.bin")
+        out = subprocess.run(
+            [cli, mdir, xbin, "50000", "10", sbin, "--labels", lbin],
+            capture_output=True, text=True, timeout=300)
+        assert out.returncode == 0, out.stderr
+        cli_scores = np.fromfile(sbin, dtype="<f4")
+        eng_scores = model.score(torch.from_numpy(X).to(dev)).cpu().numpy()
+        assert np.abs(cli_scores - eng_scores).max() < 1e-6
+        labels = np.fromfile(lbin, dtype=np.uint8)
+        expect = model.transform(torch.from_numpy(X).to(dev))[
+            "predictedLabel"].cpu().numpy()
+        np.testing.assert_array_equal(labels, expect.astype(np.uint8))
+
+    def test_cli_bf16_and_snappy(self, dev, tmp_path):
+        import subprocess
+        import os
+
+        from isolation_forest_amd.persist import model_io
+
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        cli = os.path.join(repo, "tools", "native", "ifa_score")
+        X = make_data(30000, 8, seed=52)
+        Xb = torch.from_numpy(X).to(dev).to(torch.bfloat16)
+        model = IsolationForest(numEstimators=50, randomSeed=13).fit(Xb)
+        mdir = str(tmp_path / "m")
+        model_io.save_model(model, mdir, codec="snappy")
+        xbin = str(tmp_path / "x.bin")
+        Xb.view(torch.uint16).cpu().numpy().astype("<u2").tofile(xbin)
+        sbin = str(tmp_path / "s.bin")
+        out = subprocess.run(
+            [cli, mdir, xbin, "30000", "8", sbin, "--bf16"],
+            capture_output=True, text=True, timeout=300)
+        assert out.returncode == 0, out.stderr
+        cli_scores = np.fromfile(sbin, dtype="<f4")
+        eng_scores = model.score(Xb).cpu().numpy()
+        assert np.abs(cli_scores - eng_scores).max() < 1e-6
