@@ -430,3 +430,26 @@ class TestNativeCLI:
         cli_scores = np.fromfile(sbin, dtype="<f4")
         eng_scores = model.score(Xb).cpu().numpy()
         assert np.abs(cli_scores - eng_scores).max() < 1e-6
+
+    def test_cli_extended_models(self, dev, tmp_path):
+        import subprocess
+        import os
+
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        cli = os.path.join(repo, "tools", "native", "ifa_score")
+        X = make_data(30000, 12, seed=53)
+        for ext_level, tag in [(2, "sparse"), (11, "dense")]:
+            model = ExtendedIsolationForest(
+                numEstimators=40, extensionLevel=ext_level,
+                randomSeed=14).fit(torch.from_numpy(X).to(dev))
+            mdir = str(tmp_path / f"eif_{tag}")
+            model.save(mdir)
+            xbin = str(tmp_path / f"x_{tag}.bin")
+            X.astype("<f4").tofile(xbin)
+            sbin = str(tmp_path / f"s_{tag}.bin")
+            out = subprocess.run([cli, mdir, xbin, "30000", "12", sbin],
+                                 capture_output=True, text=True, timeout=300)
+            assert out.returncode == 0, (tag, out.stderr)
+            cli_scores = np.fromfile(sbin, dtype="<f4")
+            eng = model.score(torch.from_numpy(X).to(dev)).cpu().numpy()
+            assert np.abs(cli_scores - eng).max() < 1e-6, tag

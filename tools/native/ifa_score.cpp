@@ -20,8 +20,9 @@
 //
 // Scores match the Python engine to <= 1e-6 (identical walk decisions via
 // the same integer-key packing; leaf constants differ only by libm-vs-numpy
-// float32 log ulps). EIF models are not supported here (use the Python
-// engine); the tool exits with a clear message.
+// float32 log ulps). Extended models route through the same sparse/dense
+// kernels as the Python engine (nnz <= 5 sparse, d <= 32 densified dense);
+// wide-d EIF falls back to the Python engine with a clear message.
 
 #include <hip/hip_runtime.h>
 #include <zlib.h>
@@ -48,6 +49,23 @@ void launch_score_forest(bool bf16, int rpt, bool rows_lds, int ilp,
                          int32_t max_nodes, int32_t height_limit, float fT,
                          float c_norm, int finalize, size_t lds, int blocks,
                          hipStream_t stream);
+void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
+                                    const void* nodes, const float* values,
+                                    const float* hw, const int32_t* ncount,
+                                    float* out, int64_t N, int32_t d,
+                                    int32_t T, int32_t max_nodes,
+                                    int32_t height_limit, float fT,
+                                    float c_norm, int finalize, size_t lds,
+                                    int blocks, hipStream_t stream);
+void launch_score_extended_sparse_v2(bool bf16, int nnz, const void* X,
+                                     const void* nodes, const float* values,
+                                     const int32_t* hidx, const float* hw,
+                                     const int32_t* ncount, float* out,
+                                     int64_t N, int32_t d, int32_t dpad,
+                                     int32_t T, int32_t max_nodes,
+                                     int32_t height_limit, float fT,
+                                     float c_norm, int finalize, size_t lds,
+                                     int blocks, hipStream_t stream);
 }
 
 #define HIP_CHECK(x)                                                         \
@@ -200,11 +218,14 @@ struct NodeRec {
   int32_t id, left, right, split_attr;
   double split_value;
   int64_t num_instances;
+  std::vector<int32_t> indices;  // extended only
+  std::vector<float> weights;    // extended only
 };
 
 struct Model {
   int num_samples = 0, num_features = 0;
   double threshold = -1.0;
+  bool extended = false;
   std::map<int, std::vector<NodeRec>> trees;
 };
 
@@ -224,10 +245,7 @@ static Model load_model(const std::string& dir) {
   if (metas.empty()) throw std::runtime_error("no metadata under " + dir);
   auto mb = read_file(metas[0]);
   std::string meta(mb.begin(), mb.end());
-  if (meta.find("Extended") != std::string::npos)
-    throw std::runtime_error(
-        "extended isolation forest models are not supported by ifa_score; "
-        "use the Python engine");
+  m.extended = meta.find("Extended") != std::string::npos;
   double v;
   if (find_json_number(meta, "numSamples", &v)) m.num_samples = (int)v;
   if (find_json_number(meta, "numFeatures", &v)) m.num_features = (int)v;
@@ -256,9 +274,7 @@ static Model load_model(const std::string& dir) {
         if (ks == "avro.schema") {
           std::string sch(val.begin(), val.end());
           if (sch.find("extendedNodeData") != std::string::npos)
-            throw std::runtime_error(
-                "extended isolation forest models are not supported by "
-                "ifa_score; use the Python engine");
+            m.extended = true;
         }
       }
     }
@@ -291,9 +307,36 @@ static Model load_model(const std::string& dir) {
         nd.id = (int32_t)br.zz();
         nd.left = (int32_t)br.zz();
         nd.right = (int32_t)br.zz();
-        nd.split_attr = (int32_t)br.zz();
-        nd.split_value = br.f64();
-        nd.num_instances = br.zz();
+        if (!m.extended) {
+          nd.split_attr = (int32_t)br.zz();
+          nd.split_value = br.f64();
+          nd.num_instances = br.zz();
+        } else {
+          if (br.zz() != 0)
+            throw std::runtime_error("null indices array");
+          while (true) {  // indices array blocks
+            int64_t c = br.zz();
+            if (c == 0) break;
+            if (c < 0) { br.zz(); c = -c; }
+            for (int64_t j = 0; j < c; ++j)
+              nd.indices.push_back((int32_t)br.zz());
+          }
+          if (br.zz() != 0)
+            throw std::runtime_error("null weights array");
+          while (true) {  // weights array blocks (float items)
+            int64_t c = br.zz();
+            if (c == 0) break;
+            if (c < 0) { br.zz(); c = -c; }
+            for (int64_t j = 0; j < c; ++j) {
+              float w;
+              br.bytes(&w, 4);
+              nd.weights.push_back(w);
+            }
+          }
+          nd.split_value = br.f64();  // offset
+          nd.num_instances = br.zz();
+          nd.split_attr = nd.indices.empty() ? -1 : (int32_t)nd.indices.size();
+        }
         m.trees[tree].push_back(nd);
       }
     }
@@ -401,6 +444,88 @@ static Packed pack_v4(const Model& m, int d_sentinel, bool bf16) {
 }
 
 // ---------------------------------------------------------------------------
+// EIF packing (mirror of ops/gpu_engine._eif_nodes_values/_eif_dense_packed)
+// ---------------------------------------------------------------------------
+
+struct PackedEIF {
+  std::vector<int32_t> nodes;   // [T][mn][2]: {right|self<<12, off32/-inf}
+  std::vector<float> values;    // [T][mn] depth-folded leaf values
+  std::vector<int32_t> ncount;  // [T]
+  std::vector<float> hw_dense;  // [T][mn][D] when dense route
+  std::vector<int32_t> hidx;    // [T][mn][nnz] when sparse route
+  std::vector<float> hw_sparse; // [T][mn][nnz]
+  int T = 0, mn = 0, nnz = 0, max_depth = 1;
+  bool uniform = true;
+};
+
+static PackedEIF pack_eif(const Model& m, int D_dense, bool want_sparse) {
+  PackedEIF p;
+  p.T = m.trees.empty() ? 0 : (m.trees.rbegin()->first + 1);
+  for (const auto& kv : m.trees) {
+    p.mn = std::max(p.mn, (int)kv.second.size());
+    for (const auto& nd : kv.second)
+      p.nnz = std::max(p.nnz, (int)nd.indices.size());
+  }
+  p.mn = std::max(p.mn, 1);
+  p.nnz = std::max(p.nnz, 1);
+  if (p.mn > 32767) throw std::runtime_error("forest too deep to pack");
+  p.nodes.assign((size_t)p.T * p.mn * 2, 0);
+  p.values.assign((size_t)p.T * p.mn, 0.0f);
+  p.ncount.assign(std::max(p.T, 1), 1);
+  if (want_sparse) {
+    p.hidx.assign((size_t)p.T * p.mn * p.nnz, 0);
+    p.hw_sparse.assign((size_t)p.T * p.mn * p.nnz, 0.0f);
+  } else {
+    p.hw_dense.assign((size_t)p.T * p.mn * D_dense, 0.0f);
+  }
+  const float ninf = -INFINITY;
+  std::vector<int> depth(p.mn);
+  for (const auto& kv : m.trees) {
+    int t = kv.first;
+    std::vector<NodeRec> sorted(kv.second);
+    std::sort(sorted.begin(), sorted.end(),
+              [](const NodeRec& a, const NodeRec& b) { return a.id < b.id; });
+    p.ncount[t] = (int32_t)sorted.size();
+    std::fill(depth.begin(), depth.end(), 0);
+    for (const auto& nd : sorted)
+      if (nd.left != -1) {
+        if (nd.left != nd.id + 1)
+          throw std::runtime_error("pre-order invariant broken");
+        depth.at(nd.left) = depth.at(nd.id) + 1;
+        depth.at(nd.right) = depth.at(nd.id) + 1;
+      }
+    int32_t* base = p.nodes.data() + (size_t)t * p.mn * 2;
+    float* vals = p.values.data() + (size_t)t * p.mn;
+    for (const auto& nd : sorted) {
+      if (nd.left == -1) {
+        base[2 * nd.id] = nd.id << 12;  // self-loop
+        memcpy(&base[2 * nd.id + 1], &ninf, 4);
+        vals[nd.id] =
+            (float)depth.at(nd.id) + avg_path_length32(nd.num_instances);
+        p.max_depth = std::max(p.max_depth, depth.at(nd.id));
+      } else {
+        base[2 * nd.id] = nd.right << 12;
+        float off32 = (float)nd.split_value;
+        memcpy(&base[2 * nd.id + 1], &off32, 4);
+        p.max_depth = std::max(p.max_depth, depth.at(nd.id) + 1);
+        if ((int)nd.indices.size() != p.nnz) p.uniform = false;
+        for (size_t j = 0; j < nd.indices.size(); ++j) {
+          if (want_sparse) {
+            p.hidx[((size_t)t * p.mn + nd.id) * p.nnz + j] = nd.indices[j];
+            p.hw_sparse[((size_t)t * p.mn + nd.id) * p.nnz + j] =
+                nd.weights[j];
+          } else {
+            p.hw_dense[((size_t)t * p.mn + nd.id) * D_dense +
+                       nd.indices[j]] = nd.weights[j];
+          }
+        }
+      }
+    }
+  }
+  return p;
+}
+
+// ---------------------------------------------------------------------------
 // main
 // ---------------------------------------------------------------------------
 
@@ -440,8 +565,7 @@ int main(int argc, char** argv) {
     fprintf(stderr, "error: d must be <= 4094\n");
     return 2;
   }
-  Packed p = pack_v4(m, d, bf16);
-  if (p.T == 0) {
+  if (m.trees.empty()) {
     fprintf(stderr, "error: model has no trees; cannot score\n");
     return 2;
   }
@@ -459,50 +583,132 @@ int main(int argc, char** argv) {
     return 2;
   }
 
-  void *dX, *dNodes, *dNcount, *dOut;
+  void *dX, *dOut;
   HIP_CHECK(hipMalloc(&dX, data.size()));
-  HIP_CHECK(hipMalloc(&dNodes, p.nodes.size() * 4));
-  HIP_CHECK(hipMalloc(&dNcount, p.ncount.size() * 4));
   HIP_CHECK(hipMalloc(&dOut, (size_t)N * 4));
   HIP_CHECK(hipMemcpy(dX, data.data(), data.size(), hipMemcpyHostToDevice));
-  HIP_CHECK(hipMemcpy(dNodes, p.nodes.data(), p.nodes.size() * 4,
-                      hipMemcpyHostToDevice));
-  HIP_CHECK(hipMemcpy(dNcount, p.ncount.data(), p.ncount.size() * 4,
-                      hipMemcpyHostToDevice));
 
-  // launch-config mirror of bindings.cpp::score_forest
+  const size_t kMaxLds = 160 * 1024;
   int64_t dpad = d + 1;
   if (bf16) {
     while (dpad % 4 != 2) ++dpad;
   } else {
     while (dpad % 2 != 1) ++dpad;
   }
-  const size_t kMaxLds = 160 * 1024;
-  size_t node_bytes = (size_t)4 * p.mn * 8;
-  if (node_bytes > kMaxLds) {
-    fprintf(stderr, "error: tree too large for LDS staging\n");
-    return 2;
-  }
-  int rpt = 2;
-  bool rows_lds = true;
-  size_t lds;
-  for (;;) {
-    size_t row_bytes = (size_t)rpt * 256 * dpad * elem;
-    lds = node_bytes + row_bytes;
-    if (lds * 3 <= kMaxLds || (rpt == 1 && lds <= 150 * 1024)) break;
-    if (rpt == 2) { rpt = 1; continue; }
-    rows_lds = false;
-    lds = node_bytes;
-    break;
-  }
-  int64_t rows_per_block = (int64_t)(rows_lds ? rpt : 1) * 256;
-  int blocks = (int)std::min<int64_t>(
-      (N + rows_per_block - 1) / rows_per_block, 8192);
+  int T_report = 0;
 
-  ifa::launch_score_forest(bf16, rpt, rows_lds, 4, dX, dNodes,
-                           (const int32_t*)dNcount, (float*)dOut, N, d,
-                           (int32_t)dpad, p.Tpad, p.mn, p.max_depth,
-                           (float)p.T, c_norm, 1, lds, blocks, 0);
+  if (!m.extended) {
+    Packed p = pack_v4(m, d, bf16);
+    T_report = p.T;
+    void *dNodes, *dNcount;
+    HIP_CHECK(hipMalloc(&dNodes, p.nodes.size() * 4));
+    HIP_CHECK(hipMalloc(&dNcount, p.ncount.size() * 4));
+    HIP_CHECK(hipMemcpy(dNodes, p.nodes.data(), p.nodes.size() * 4,
+                        hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(dNcount, p.ncount.data(), p.ncount.size() * 4,
+                        hipMemcpyHostToDevice));
+    size_t node_bytes = (size_t)4 * p.mn * 8;
+    if (node_bytes > kMaxLds) {
+      fprintf(stderr, "error: tree too large for LDS staging\n");
+      return 2;
+    }
+    int rpt = 2;
+    bool rows_lds = true;
+    size_t lds;
+    for (;;) {
+      size_t row_bytes = (size_t)rpt * 256 * dpad * elem;
+      lds = node_bytes + row_bytes;
+      if (lds * 3 <= kMaxLds || (rpt == 1 && lds <= 150 * 1024)) break;
+      if (rpt == 2) { rpt = 1; continue; }
+      rows_lds = false;
+      lds = node_bytes;
+      break;
+    }
+    int64_t rows_per_block = (int64_t)(rows_lds ? rpt : 1) * 256;
+    int blocks = (int)std::min<int64_t>(
+        (N + rows_per_block - 1) / rows_per_block, 8192);
+    ifa::launch_score_forest(bf16, rpt, rows_lds, 4, dX, dNodes,
+                             (const int32_t*)dNcount, (float*)dOut, N, d,
+                             (int32_t)dpad, p.Tpad, p.mn, p.max_depth,
+                             (float)p.T, c_norm, 1, lds, blocks, 0);
+  } else {
+    // routing mirror of ops/gpu_engine.score_extended_forest (the wide-d
+    // general kernel is Python-only; practical configs are covered here)
+    int nnz_seen = 1;
+    for (const auto& kv : m.trees)
+      for (const auto& nd : kv.second)
+        nnz_seen = std::max(nnz_seen, (int)nd.indices.size());
+    int blocks = (int)std::min<int64_t>((N + 511) / 512, 8192);
+    bool try_sparse = nnz_seen <= 5;
+    if (try_sparse) {
+      PackedEIF p = pack_eif(m, 0, /*want_sparse=*/true);
+      size_t lds = (size_t)p.mn * (12 + p.nnz * 8) +
+                   (size_t)2 * 256 * dpad * elem;
+      if (p.uniform && lds <= 150 * 1024) {
+        T_report = p.T;
+        void *dNodes, *dVals, *dHidx, *dHw, *dNcount;
+        HIP_CHECK(hipMalloc(&dNodes, p.nodes.size() * 4));
+        HIP_CHECK(hipMalloc(&dVals, p.values.size() * 4));
+        HIP_CHECK(hipMalloc(&dHidx, p.hidx.size() * 4));
+        HIP_CHECK(hipMalloc(&dHw, p.hw_sparse.size() * 4));
+        HIP_CHECK(hipMalloc(&dNcount, p.ncount.size() * 4));
+        HIP_CHECK(hipMemcpy(dNodes, p.nodes.data(), p.nodes.size() * 4,
+                            hipMemcpyHostToDevice));
+        HIP_CHECK(hipMemcpy(dVals, p.values.data(), p.values.size() * 4,
+                            hipMemcpyHostToDevice));
+        HIP_CHECK(hipMemcpy(dHidx, p.hidx.data(), p.hidx.size() * 4,
+                            hipMemcpyHostToDevice));
+        HIP_CHECK(hipMemcpy(dHw, p.hw_sparse.data(), p.hw_sparse.size() * 4,
+                            hipMemcpyHostToDevice));
+        HIP_CHECK(hipMemcpy(dNcount, p.ncount.data(), p.ncount.size() * 4,
+                            hipMemcpyHostToDevice));
+        int64_t dpad_s = d;  // sparse kernel stride (no sentinel column)
+        if (bf16) { while (dpad_s % 4 != 2) ++dpad_s; }
+        else { while (dpad_s % 2 != 1) ++dpad_s; }
+        ifa::launch_score_extended_sparse_v2(
+            bf16, p.nnz, dX, dNodes, (const float*)dVals,
+            (const int32_t*)dHidx, (const float*)dHw,
+            (const int32_t*)dNcount, (float*)dOut, N, d, (int32_t)dpad_s,
+            p.T, p.mn, p.max_depth, (float)p.T, c_norm, 1, lds, blocks, 0);
+        goto launched;
+      }
+      try_sparse = false;
+    }
+    if (!try_sparse) {
+      if (d > 32) {
+        fprintf(stderr,
+                "error: extended models with d > 32 and nnz > 5 are not "
+                "supported by ifa_score; use the Python engine\n");
+        return 2;
+      }
+      int D = d <= 8 ? 8 : (d <= 16 ? 16 : 32);
+      PackedEIF p = pack_eif(m, D, /*want_sparse=*/false);
+      T_report = p.T;
+      size_t lds = (size_t)p.mn * 12 + 16 + (size_t)p.mn * (D / 4 + 1) * 16;
+      if (lds > kMaxLds) {
+        fprintf(stderr, "error: tree too large for LDS staging\n");
+        return 2;
+      }
+      void *dNodes, *dVals, *dHw, *dNcount;
+      HIP_CHECK(hipMalloc(&dNodes, p.nodes.size() * 4));
+      HIP_CHECK(hipMalloc(&dVals, p.values.size() * 4));
+      HIP_CHECK(hipMalloc(&dHw, p.hw_dense.size() * 4));
+      HIP_CHECK(hipMalloc(&dNcount, p.ncount.size() * 4));
+      HIP_CHECK(hipMemcpy(dNodes, p.nodes.data(), p.nodes.size() * 4,
+                          hipMemcpyHostToDevice));
+      HIP_CHECK(hipMemcpy(dVals, p.values.data(), p.values.size() * 4,
+                          hipMemcpyHostToDevice));
+      HIP_CHECK(hipMemcpy(dHw, p.hw_dense.data(), p.hw_dense.size() * 4,
+                          hipMemcpyHostToDevice));
+      HIP_CHECK(hipMemcpy(dNcount, p.ncount.data(), p.ncount.size() * 4,
+                          hipMemcpyHostToDevice));
+      ifa::launch_score_extended_dense_v2(
+          bf16, D, dX, dNodes, (const float*)dVals, (const float*)dHw,
+          (const int32_t*)dNcount, (float*)dOut, N, d, p.T, p.mn,
+          p.max_depth, (float)p.T, c_norm, 1, lds, blocks, 0);
+    }
+  }
+launched:
   HIP_CHECK(hipGetLastError());
   HIP_CHECK(hipDeviceSynchronize());
 
@@ -530,8 +736,8 @@ int main(int argc, char** argv) {
     for (int64_t i = 0; i < N; ++i)
       flagged += (double)scores[i] >= m.threshold;
   fprintf(stderr,
-          "scored %lld rows x %d trees (%s); threshold=%.6f flagged=%lld\n",
-          (long long)N, p.T, bf16 ? "bf16" : "f32", m.threshold,
-          (long long)flagged);
+          "scored %lld rows x %d trees (%s%s); threshold=%.6f flagged=%lld\n",
+          (long long)N, T_report, m.extended ? "extended, " : "",
+          bf16 ? "bf16" : "f32", m.threshold, (long long)flagged);
   return 0;
 }
