@@ -1,24 +1,37 @@
 // MI355X (gfx950, CDNA4) Isolation Forest kernels.
 //
-// These are the hand-written HIP implementations of the reference's JVM hot
-// loops:
+// Hand-written HIP implementations of the reference's JVM hot loops
+// (design rationale and measured limits in
+// docs/DESIGN.md):
 //  * build_forest_kernel          — K1 (min/max scan + constant-feature
 //    retry), K2 (stable row partition), K11 (per-node feature Fisher-Yates):
 //    one 64-lane wavefront per tree, row-index permutation in LDS, ballot/
-//    popcount stable partition, shfl-xor min/max reductions.
+//    popcount stable partition, shfl-xor min/max reductions; emits per-node
+//    depths for the scoring pack.
 //  * build_extended_forest_kernel — K3-K5 (EIF Gaussian hyperplane draw,
 //    L2 normalize, per-coordinate min/max + offset, dot-product partition).
-//  * score_forest_kernel          — K6/K8 (batched path-length traversal):
-//    one thread per row, the current tree's nodes staged in LDS as 16-byte
-//    AoS records (one ds_read_b128 per visit), rows optionally staged in
-//    LDS (padded to kill bank conflicts), float32 path-sum accumulation in
-//    tree order (bitwise == the CPU oracle), leaf c(n) terms precomputed.
-//  * score_extended_forest_kernel — K7 (EIF traversal, sparse dot per level).
+//  * score_forest_v4              — K6/K8 standard scoring: integer-KEY
+//    compares (rows staged in LDS as order-preserving u16/u32 keys,
+//    per-node integer thresholds), depth folded into f32 leaf values,
+//    self-looping leaves => FIXED-trip walks with zero bookkeeping,
+//    2 rows x 4 staged trees = 8 chains/thread with batch-phased LDS reads
+//    (staggered lgkmcnt waits). Bitwise == cpu_engine.path_lengths.
+//  * score_extended_dense_v2      — K7 fast path (hyperplanes densified to
+//    D in {8,16,32} columns): rows resident in f32 REGISTERS across the
+//    whole tree loop, one tree's nodes+values+weights staged in LDS
+//    (conflict-breaking D/4+1 float4 row stride), exec-masked finished
+//    lanes, offset=-inf self-looping leaves. Tolerance contract (4-partial
+//    fma dot), routing in ops/gpu_engine.score_extended_forest.
+//  * score_extended_sparse_v2     — K7 small-nnz path (templated NNZ<=5):
+//    same fixed-trip structure, strict oracle j-order dot (bitwise).
+//  * score_extended_forest_kernel — K7 general strict-order fallback
+//    (wide d / ragged hyperplane widths).
 //  * bag_gather_kernel            — K9/K10 device side: gather sampled rows
 //    into per-tree bags (indices drawn host-side by the same Philox).
 //
 // PRECISION CONTRACT: see core/cpu_engine.py. All randomness: philox.h;
 // EIF gaussians: det_math.h. Wavefront size is 64 (CDNA4) and is hard-coded.
+// This TU is torch-free: tools/native/ifa_score.cpp links it directly.
 
 #include <hip/hip_runtime.h>
 #include <stdint.h>
