@@ -1,10 +1,18 @@
 """GPU engine orchestration: device builds and scoring via the HIP extension.
 
 The canonical forest lives in host numpy (core/forest.py) — it is small
-(MBs) next to the data (GBs-TBs); device copies are cached per model and
-device. Bags/feature subsets are drawn host-side with the same Philox
-counters the kernels use, so a GPU-built forest is BIT-IDENTICAL to the
-CPU oracle's for the same seed (tests/test_gpu.py enforces this).
+(MBs) next to the data (GBs-TBs). Bags/feature subsets are drawn host-side
+with the same Philox counters the kernels use, so a GPU-built forest is
+BIT-IDENTICAL to the CPU oracle's for the same seed (tests/test_gpu.py
+enforces this).
+
+Scoring packs (the v4 integer-key node format, EIF dense/sparse variants)
+are built ON DEVICE with torch ops from the raw build outputs (depths come
+from the build kernels) and cached per model+device; the numpy host pack
+is the fallback for loaded models and is bitwise-identical
+(tests/test_gpu.py::TestDevicePacking). EIF scoring routes by hyperplane
+width: nnz<=5 uniform -> sparse v2 (bitwise), else d<=32 -> densified
+dense v2 (tolerance), else the general strict-order kernel.
 """
 
 from __future__ import annotations
