@@ -1,0 +1,136 @@
+#!/usr/bin/env python3
+"""Differential GPU<->CPU fuzzer.
+
+Random configurations of (rows, features, subsampling, tree count,
+extension level, dtype, bootstrap, data pathologies): build on the GPU and
+on the CPU oracle, demand BITWISE forest equality; score on both, demand
+the routing contract (bitwise for standard/sparse walks, knife-edge
+tolerance for reassociated dense dots). Exits non-zero on the first
+divergence with a repro line.
+
+    python tools/fuzz_parity.py [--iters 30] [--seed 0]
+"""
+import argparse
+import os
+import sys
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+from isolation_forest_amd.core import cpu_engine
+from isolation_forest_amd.ops import gpu_engine
+from isolation_forest_amd.utils.params import ResolvedParams
+
+
+def one_case(rs: np.random.RandomState, it: int) -> str:
+    rows = int(rs.randint(300, 20000))
+    d = int(rs.choice([1, 2, 3, 5, 8, 13, 16, 31, 32, 47, 64]))
+    n = int(rs.choice([2, 4, 16, 64, 128, 256, 512, 1024]))
+    n = min(n, rows)
+    k = int(rs.randint(1, d + 1))
+    T = int(rs.choice([1, 2, 3, 7, 8, 16, 33]))
+    seed = int(rs.randint(1, 2**30))
+    extended = bool(rs.randint(0, 2))
+    bf16 = bool(rs.randint(0, 2))
+    bootstrap = bool(rs.randint(0, 2))
+    patho = rs.choice(["normal", "const_col", "dups", "big", "tiny"])
+
+    X = rs.normal(size=(rows, d)).astype(np.float32)
+    if patho == "const_col":
+        X[:, rs.randint(0, d)] = float(rs.normal())
+    elif patho == "dups":
+        X = X[rs.randint(0, max(rows // 7, 1), size=rows)]
+    elif patho == "big":
+        X *= 1e20
+    elif patho == "tiny":
+        X *= 1e-20
+    desc = (f"it={it} rows={rows} d={d} n={n} k={k} T={T} seed={seed} "
+            f"ext={extended} bf16={bf16} boot={bootstrap} patho={patho}")
+
+    Xt = torch.from_numpy(X).to("cuda")
+    if bf16:
+        Xt = Xt.to(torch.bfloat16)
+        X = Xt.float().cpu().numpy()
+
+    bag = cpu_engine.sample_bags(rows, T, n, seed=seed, bootstrap=bootstrap)
+    fs = cpu_engine.feature_subsets(d, k, T, seed=seed)
+
+    if not extended:
+        cpu_f = cpu_engine.build_forest(X, bag, fs, seed, n, k, d)
+        rp = ResolvedParams(num_samples=n, num_features=k, total_rows=rows,
+                            total_features=d)
+        gpu_f = gpu_engine.build_forest(Xt, bag, fs, seed, rp)
+        for name in ["node_count", "feature", "right", "num_instances"]:
+            if not np.array_equal(getattr(gpu_f, name), getattr(cpu_f, name)):
+                return f"FOREST MISMATCH {name}: {desc}"
+        if not np.array_equal(gpu_f.value.view(np.int32),
+                              cpu_f.value.view(np.int32)):
+            return f"FOREST MISMATCH value: {desc}"
+        cpu_ps = cpu_engine.path_lengths(cpu_f, X)
+
+        class Shell:
+            forest = gpu_f
+            _gpu_forest_cache = {}
+        gpu_ps = gpu_engine.score_forest(Shell(), Xt, finalize=False)
+        if not np.array_equal(gpu_ps.cpu().numpy().view(np.int32),
+                              cpu_ps.view(np.int32)):
+            return f"SCORE MISMATCH standard (bitwise): {desc}"
+    else:
+        ext_level = int(rs.randint(0, k))
+        desc += f" ext_level={ext_level}"
+        cpu_f = cpu_engine.build_extended_forest(X, bag, fs, seed, n, k, d,
+                                                 ext_level)
+        rp = ResolvedParams(num_samples=n, num_features=k, total_rows=rows,
+                            total_features=d, extension_level=ext_level)
+        gpu_f = gpu_engine.build_extended_forest(Xt, bag, fs, seed, rp)
+        for name in ["node_count", "feature", "right", "num_instances",
+                     "hyper_idx"]:
+            if not np.array_equal(getattr(gpu_f, name), getattr(cpu_f, name)):
+                return f"EIF FOREST MISMATCH {name}: {desc}"
+        for name in ["value", "hyper_w"]:
+            if not np.array_equal(getattr(gpu_f, name).view(np.int32),
+                                  getattr(cpu_f, name).view(np.int32)):
+                return f"EIF FOREST MISMATCH {name}: {desc}"
+        if not np.array_equal(gpu_f.offset64, cpu_f.offset64):
+            return f"EIF FOREST MISMATCH offset64: {desc}"
+        cpu_ps = cpu_engine.path_lengths_extended(cpu_f, X)
+
+        class Shell:
+            forest = gpu_f
+            _gpu_forest_cache = {}
+        gpu_ps = gpu_engine.score_extended_forest(Shell(), Xt, finalize=False)
+        diff = np.abs(gpu_ps.cpu().numpy() - cpu_ps)
+        nnz = min(ext_level + 1, k)
+        if nnz <= 5:  # sparse v2 or general: strict j-order, bitwise
+            if not np.array_equal(gpu_ps.cpu().numpy().view(np.int32),
+                                  cpu_ps.view(np.int32)):
+                return f"SCORE MISMATCH EIF sparse (bitwise): {desc}"
+        else:  # dense: reassociated dot, knife-edge tolerance
+            frac = float((diff > 1e-3 * max(1.0, np.abs(cpu_ps).max())).mean())
+            if frac > 0.01:
+                return (f"SCORE MISMATCH EIF dense (>{frac:.4f} rows off): "
+                        f"{desc}")
+    return ""
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--iters", type=int, default=30)
+    ap.add_argument("--seed", type=int, default=0)
+    args = ap.parse_args()
+    from isolation_forest_amd.ops import load_extension
+
+    load_extension()
+    rs = np.random.RandomState(args.seed)
+    for it in range(args.iters):
+        msg = one_case(rs, it)
+        if msg:
+            print(msg)
+            sys.exit(1)
+    print(f"fuzz: {args.iters} random configurations, all parity checks "
+          "passed")
+
+
+if __name__ == "__main__":
+    main()
