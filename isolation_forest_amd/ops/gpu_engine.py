@@ -610,12 +610,17 @@ def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torc
             )
     if d <= 32 and (nnz == d or nnz >= 6):
         D = 8 if d <= 8 else (16 if d <= 16 else 32)
-        aos, ncount, extra = _device_forest(
-            model, X.device, v4_key=("eif_dense", D))
-        return ext.score_extended_dense_v2(
-            X.contiguous(), aos, extra["values"], extra["hw"], ncount,
-            extra["height"], c, finalize,
-        )
+        mn = forest.feature.shape[1]
+        lds = mn * 12 + 16 + mn * (D // 4 + 1) * 16
+        if lds <= 160 * 1024:
+            aos, ncount, extra = _device_forest(
+                model, X.device, v4_key=("eif_dense", D))
+            return ext.score_extended_dense_v2(
+                X.contiguous(), aos, extra["values"], extra["hw"], ncount,
+                extra["height"], c, finalize,
+            )
+        # deep forests overflow the dense kernel's LDS weight staging:
+        # fall through to the general kernel (nodes from global if needed)
     aos, ncount, extra = _device_forest(model, X.device)
     return ext.score_extended_forest(
         X.contiguous(), aos, extra["hidx"], extra["hw"], ncount, c, finalize

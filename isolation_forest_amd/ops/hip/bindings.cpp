@@ -33,8 +33,8 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
                                   int32_t max_nodes, int32_t height_limit,
                                   size_t lds, hipStream_t stream);
 
-void launch_score_forest(bool bf16, int rpt, bool rows_lds, int ilp,
-                         const void* X, const void* nodes,
+void launch_score_forest(bool bf16, int rpt, bool rows_lds, bool nodes_lds,
+                         int ilp, const void* X, const void* nodes,
                          const int32_t* ncount, float* out, int64_t N,
                          int32_t d, int32_t dpad, int32_t Tpad,
                          int32_t max_nodes, int32_t height_limit, float fT,
@@ -50,13 +50,13 @@ void launch_score_extended_dense(bool bf16, bool rows_lds, bool wlds,
                                  size_t lds, int blocks, hipStream_t stream);
 
 void launch_score_extended_forest(bool bf16, bool rows_lds, bool hyper_lds,
-                                  const void* X, const void* nodes,
-                                  const int32_t* hidx, const float* hw,
-                                  const int32_t* ncount, float* out, int64_t N,
-                                  int32_t d, int32_t T, int32_t max_nodes,
-                                  int32_t nnz, float fT, float c_norm,
-                                  int finalize, size_t lds, int blocks,
-                                  hipStream_t stream);
+                                  bool nodes_lds, const void* X,
+                                  const void* nodes, const int32_t* hidx,
+                                  const float* hw, const int32_t* ncount,
+                                  float* out, int64_t N, int32_t d, int32_t T,
+                                  int32_t max_nodes, int32_t nnz, float fT,
+                                  float c_norm, int finalize, size_t lds,
+                                  int blocks, hipStream_t stream);
 
 void launch_score_extended_sparse_v2(bool bf16, int nnz, const void* X,
                                      const void* nodes, const float* values,
@@ -239,8 +239,14 @@ torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
   } else {
     while (dpad % 2 != 1) ++dpad;  // odd word count
   }
-  const size_t node_bytes = (size_t)ilp * max_nodes * 8;  // staged trees
-  TORCH_CHECK(node_bytes <= kMaxLds, "tree too large for LDS staging");
+  // deep forests (large maxSamples) overflow LDS node staging: walk nodes
+  // from global/L2 instead (NODES_LDS=false) — correct for any tree size
+  size_t node_bytes = (size_t)ilp * max_nodes * 8;  // staged trees
+  bool nodes_lds = true;
+  if (node_bytes + (size_t)256 * dpad * elem > 150 * 1024) {
+    nodes_lds = false;
+    node_bytes = 0;
+  }
   // prefer 2 rows/thread when 3 blocks/CU still fit, else 1, else global X
   int rpt = 2;
   bool rows_lds = true;
@@ -260,7 +266,7 @@ torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
   int64_t rows_per_block = (int64_t)(rows_lds ? rpt : 1) * 256;
   int blocks = (int)std::min<int64_t>(
       (N + rows_per_block - 1) / rows_per_block, 8192);
-  ifa::launch_score_forest(bf16, rpt, rows_lds, ilp, X.data_ptr(),
+  ifa::launch_score_forest(bf16, rpt, rows_lds, nodes_lds, ilp, X.data_ptr(),
                            nodes_packed.data_ptr<int32_t>(),
                            ncount.data_ptr<int32_t>(), out.data_ptr<float>(),
                            N, (int32_t)d, (int32_t)dpad, (int32_t)Tpad,
@@ -300,9 +306,8 @@ torch::Tensor score_extended_forest(torch::Tensor X,
   size_t row_bytes = (size_t)256 * (d + pad) * elem;
   size_t node_bytes = (size_t)max_nodes * 8;
   int blocks = (int)std::min<int64_t>((N + 255) / 256, 8192);
-  TORCH_CHECK(node_bytes <= kMaxLds, "tree too large for LDS staging");
 
-  if (nnz == d && (d % 4) == 0) {
+  if (nnz == d && (d % 4) == 0 && node_bytes <= 120 * 1024) {
     // fully-extended dense hyperplanes: implicit indices, vectorized dot.
     // 8-B-aligned row stride for the uint2/float4 row reads.
     int64_t pad4 = (4 - (d % 4)) % 4;
@@ -322,14 +327,17 @@ torch::Tensor score_extended_forest(torch::Tensor X,
     return out;
   }
 
+  // deep forests: stage what fits, walk the rest from global/L2
+  bool nodes_lds = node_bytes <= 120 * 1024;
+  size_t nb = nodes_lds ? node_bytes : 0;
   size_t hyper_bytes = (size_t)max_nodes * nnz * 8;
-  bool hyper_lds = node_bytes + hyper_bytes <= 96 * 1024;
+  bool hyper_lds = nb + hyper_bytes <= 96 * 1024;
   bool rows_lds =
-      node_bytes + (hyper_lds ? hyper_bytes : 0) + row_bytes <= 144 * 1024;
+      nb + (hyper_lds ? hyper_bytes : 0) + row_bytes <= 144 * 1024;
   size_t lds =
-      node_bytes + (hyper_lds ? hyper_bytes : 0) + (rows_lds ? row_bytes : 0);
+      nb + (hyper_lds ? hyper_bytes : 0) + (rows_lds ? row_bytes : 0);
   ifa::launch_score_extended_forest(
-      bf16, rows_lds, hyper_lds, X.data_ptr(),
+      bf16, rows_lds, hyper_lds, nodes_lds, X.data_ptr(),
       nodes_packed.data_ptr<int32_t>(), hidx.data_ptr<int32_t>(),
       hw.data_ptr<float>(), ncount.data_ptr<int32_t>(), out.data_ptr<float>(),
       N, (int32_t)d, (int32_t)T, (int32_t)max_nodes, (int32_t)nnz, (float)T,

@@ -529,7 +529,7 @@ __device__ __host__ __forceinline__ int row_stride(int d) {
 }
 
 
-template <typename KT, int RPT, bool ROWS_LDS, int V4_ILP>
+template <typename KT, int RPT, bool ROWS_LDS, int V4_ILP, bool NODES_LDS>
 __global__ void __launch_bounds__(256) score_forest_v4(
     const KT* __restrict__ X,          // raw bf16/f32 bits [N][d]
     const int2* __restrict__ nodes,    // [Tpad][max_nodes] packed v4
@@ -540,8 +540,8 @@ __global__ void __launch_bounds__(256) score_forest_v4(
   const int tid = threadIdx.x;
   const int rows_per_iter = RPT * 256;
 
-  int2* tlds = (int2*)smem;                    // [V4_ILP * max_nodes]
-  KT* rows = (KT*)(tlds + V4_ILP * max_nodes);  // [rows_per_iter][dpad]
+  int2* tlds = (int2*)smem;  // [V4_ILP * max_nodes] when NODES_LDS
+  KT* rows = (KT*)(tlds + (NODES_LDS ? V4_ILP * max_nodes : 0));
 
   for (int64_t block_row0 = (int64_t)blockIdx.x * rows_per_iter; block_row0 < N;
        block_row0 += (int64_t)gridDim.x * rows_per_iter) {
@@ -568,14 +568,21 @@ __global__ void __launch_bounds__(256) score_forest_v4(
     for (int r = 0; r < RPT; ++r) psum[r] = 0.f;
 
     for (int t = 0; t < Tpad; t += V4_ILP) {
-      __syncthreads();
+      if (NODES_LDS) {
+        __syncthreads();
 #pragma unroll
-      for (int s = 0; s < V4_ILP; ++s) {
-        const int ncs = ncnt[t + s];
-        const int2* ss = nodes + (int64_t)(t + s) * max_nodes;
-        for (int i = tid; i < ncs; i += 256) tlds[s * max_nodes + i] = ss[i];
+        for (int s = 0; s < V4_ILP; ++s) {
+          const int ncs = ncnt[t + s];
+          const int2* ss = nodes + (int64_t)(t + s) * max_nodes;
+          for (int i = tid; i < ncs; i += 256) tlds[s * max_nodes + i] = ss[i];
+        }
+        __syncthreads();
       }
-      __syncthreads();
+      const int2* nbase[V4_ILP];
+#pragma unroll
+      for (int s = 0; s < V4_ILP; ++s)
+        nbase[s] = NODES_LDS ? nullptr
+                             : nodes + (int64_t)(t + s) * max_nodes;
 
       int cur[RPT][V4_ILP];
 #pragma unroll
@@ -593,7 +600,8 @@ __global__ void __launch_bounds__(256) score_forest_v4(
         for (int r = 0; r < RPT; ++r)
 #pragma unroll
           for (int s = 0; s < V4_ILP; ++s)
-            nd[r][s] = tlds[s * max_nodes + cur[r][s]];
+            nd[r][s] = NODES_LDS ? tlds[s * max_nodes + cur[r][s]]
+                                 : nbase[s][cur[r][s]];
         KT kraw[RPT][V4_ILP];
 #pragma unroll
         for (int r = 0; r < RPT; ++r) {
@@ -634,7 +642,8 @@ __global__ void __launch_bounds__(256) score_forest_v4(
         for (int s = 0; s < V4_ILP; ++s)
           psum[r] = __fadd_rn(
               psum[r],
-              __int_as_float(tlds[s * max_nodes + cur[r][s]].y));
+              __int_as_float((NODES_LDS ? tlds[s * max_nodes + cur[r][s]]
+                                        : nbase[s][cur[r][s]]).y));
     }
 
 #pragma unroll
@@ -1059,7 +1068,7 @@ __global__ void __launch_bounds__(256) score_extended_sparse_v2(
 // extended scoring, sparse general path (nnz < d): exact oracle order.
 // ---------------------------------------------------------------------------
 
-template <typename XT, bool ROWS_LDS, bool HYPER_LDS>
+template <typename XT, bool ROWS_LDS, bool HYPER_LDS, bool NODES_LDS>
 __global__ void __launch_bounds__(256) score_extended_forest_kernel(
     const XT* __restrict__ X, const int2* __restrict__ nodes,
     const int32_t* __restrict__ hidx_g,  // [T][max_nodes][nnz]
@@ -1070,8 +1079,8 @@ __global__ void __launch_bounds__(256) score_extended_forest_kernel(
   const int tid = threadIdx.x;
   const int dpad = row_stride<XT>(d);
 
-  int2* tlds = (int2*)smem;                          // [max_nodes]
-  int32_t* hidx_lds = (int32_t*)(tlds + max_nodes);  // [max_nodes*nnz]
+  int2* tlds = (int2*)smem;  // [max_nodes] when NODES_LDS
+  int32_t* hidx_lds = (int32_t*)(tlds + (NODES_LDS ? max_nodes : 0));
   float* hw_lds = (float*)(hidx_lds + (HYPER_LDS ? max_nodes * nnz : 0));
   XT* rows = (XT*)(hw_lds + (HYPER_LDS ? max_nodes * nnz : 0));
 
@@ -1096,7 +1105,8 @@ __global__ void __launch_bounds__(256) score_extended_forest_kernel(
       __syncthreads();
       const int nc = ncnt[t];
       const int2* src = nodes + (int64_t)t * max_nodes;
-      for (int i = tid; i < nc; i += 256) tlds[i] = src[i];
+      if (NODES_LDS)
+        for (int i = tid; i < nc; i += 256) tlds[i] = src[i];
       if (HYPER_LDS) {
         const int64_t hbase = (int64_t)t * max_nodes * nnz;
         for (int i = tid; i < nc * nnz; i += 256) {
@@ -1111,7 +1121,7 @@ __global__ void __launch_bounds__(256) score_extended_forest_kernel(
         int cur = 0, dep = 0;
         float leaf = 0.f;
         while (true) {
-          const int2 nd = tlds[cur];
+          const int2 nd = NODES_LDS ? tlds[cur] : src[cur];
           if (nd.x < 0) {
             leaf = __int_as_float(nd.y);
             break;
@@ -1208,8 +1218,8 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
                      n, d, k, nnz, max_nodes, height_limit);
 }
 
-void launch_score_forest(bool bf16, int rpt, bool rows_lds, int ilp,
-                         const void* X, const void* nodes,
+void launch_score_forest(bool bf16, int rpt, bool rows_lds, bool nodes_lds,
+                         int ilp, const void* X, const void* nodes,
                          const int32_t* ncount, float* out, int64_t N,
                          int32_t d, int32_t dpad, int32_t Tpad,
                          int32_t max_nodes, int32_t height_limit, float fT,
@@ -1217,11 +1227,19 @@ void launch_score_forest(bool bf16, int rpt, bool rows_lds, int ilp,
                          hipStream_t stream) {
 #define LS(KT, RPT, RL, TI)                                                   \
   do {                                                                        \
-    raise_lds((const void*)score_forest_v4<KT, RPT, RL, TI>, lds);            \
-    hipLaunchKernelGGL((score_forest_v4<KT, RPT, RL, TI>), dim3(blocks),      \
-                       dim3(256), lds, stream, (const KT*)X,                  \
-                       (const int2*)nodes, ncount, out, N, d, dpad, Tpad,     \
-                       max_nodes, height_limit, fT, c_norm, finalize);        \
+    if (nodes_lds) {                                                          \
+      raise_lds((const void*)score_forest_v4<KT, RPT, RL, TI, true>, lds);    \
+      hipLaunchKernelGGL((score_forest_v4<KT, RPT, RL, TI, true>),            \
+                         dim3(blocks), dim3(256), lds, stream, (const KT*)X,  \
+                         (const int2*)nodes, ncount, out, N, d, dpad, Tpad,   \
+                         max_nodes, height_limit, fT, c_norm, finalize);      \
+    } else {                                                                  \
+      raise_lds((const void*)score_forest_v4<KT, RPT, RL, TI, false>, lds);   \
+      hipLaunchKernelGGL((score_forest_v4<KT, RPT, RL, TI, false>),           \
+                         dim3(blocks), dim3(256), lds, stream, (const KT*)X,  \
+                         (const int2*)nodes, ncount, out, N, d, dpad, Tpad,   \
+                         max_nodes, height_limit, fT, c_norm, finalize);      \
+    }                                                                         \
   } while (0)
 #define LS_ILP(KT, RPT, RL)                                                   \
   do {                                                                        \
@@ -1333,20 +1351,30 @@ void launch_score_extended_sparse_v2(bool bf16, int nnz, const void* X,
 }
 
 void launch_score_extended_forest(bool bf16, bool rows_lds, bool hyper_lds,
-                                  const void* X, const void* nodes,
-                                  const int32_t* hidx, const float* hw,
-                                  const int32_t* ncount, float* out, int64_t N,
-                                  int32_t d, int32_t T, int32_t max_nodes,
-                                  int32_t nnz, float fT, float c_norm,
-                                  int finalize, size_t lds, int blocks,
-                                  hipStream_t stream) {
+                                  bool nodes_lds, const void* X,
+                                  const void* nodes, const int32_t* hidx,
+                                  const float* hw, const int32_t* ncount,
+                                  float* out, int64_t N, int32_t d, int32_t T,
+                                  int32_t max_nodes, int32_t nnz, float fT,
+                                  float c_norm, int finalize, size_t lds,
+                                  int blocks, hipStream_t stream) {
 #define LSE(XT, RL, HL)                                                       \
   do {                                                                        \
-    raise_lds((const void*)score_extended_forest_kernel<XT, RL, HL>, lds);    \
-    hipLaunchKernelGGL((score_extended_forest_kernel<XT, RL, HL>),            \
-                       dim3(blocks), dim3(256), lds, stream, (const XT*)X,    \
-                       (const int2*)nodes, hidx, hw, ncount, out, N, d, T,    \
-                       max_nodes, nnz, fT, c_norm, finalize);                 \
+    if (nodes_lds) {                                                          \
+      raise_lds((const void*)score_extended_forest_kernel<XT, RL, HL, true>,  \
+                lds);                                                         \
+      hipLaunchKernelGGL((score_extended_forest_kernel<XT, RL, HL, true>),    \
+                         dim3(blocks), dim3(256), lds, stream, (const XT*)X,  \
+                         (const int2*)nodes, hidx, hw, ncount, out, N, d, T,  \
+                         max_nodes, nnz, fT, c_norm, finalize);               \
+    } else {                                                                  \
+      raise_lds((const void*)score_extended_forest_kernel<XT, RL, HL, false>, \
+                lds);                                                         \
+      hipLaunchKernelGGL((score_extended_forest_kernel<XT, RL, HL, false>),   \
+                         dim3(blocks), dim3(256), lds, stream, (const XT*)X,  \
+                         (const int2*)nodes, hidx, hw, ncount, out, N, d, T,  \
+                         max_nodes, nnz, fT, c_norm, finalize);               \
+    }                                                                         \
   } while (0)
   if (bf16) {
     if (rows_lds && hyper_lds) LSE(uint16_t, true, true);

@@ -453,3 +453,43 @@ class TestNativeCLI:
             cli_scores = np.fromfile(sbin, dtype="<f4")
             eng = model.score(torch.from_numpy(X).to(dev)).cpu().numpy()
             assert np.abs(cli_scores - eng).max() < 1e-6, tag
+
+
+class TestDeepForests:
+    """Large maxSamples => trees too deep for LDS node staging: the
+    nodes-from-global kernel variants must stay bitwise."""
+
+    def test_standard_max_samples_4096(self, dev):
+        from isolation_forest_amd.ops import gpu_engine
+
+        X = make_data(9000, 6, seed=61)
+        bag = cpu_engine.sample_bags(9000, 4, 4096, seed=15, bootstrap=False)
+        fs = cpu_engine.feature_subsets(6, 6, 4, seed=15)
+        forest = cpu_engine.build_forest(X, bag, fs, 15, 4096, 6, 6)
+        cpu_ps = cpu_engine.path_lengths(forest, X)
+        model = IsolationForest(numEstimators=4).fit(X[:500])
+        model.forest = forest
+        model._gpu_forest_cache = {}
+        gpu_ps = gpu_engine.score_forest(
+            model, torch.from_numpy(X).to(dev), finalize=False)
+        np.testing.assert_array_equal(
+            gpu_ps.cpu().numpy().view(np.int32), cpu_ps.view(np.int32))
+
+    def test_extended_max_samples_2048_full_ext(self, dev):
+        X = torch.from_numpy(make_data(12000, 6, seed=62)).to(dev)
+        model = ExtendedIsolationForest(
+            numEstimators=8, maxSamples=2048.0, randomSeed=16).fit(X)
+        s_gpu = model.score(X).cpu().numpy()
+        s_cpu = model.score(X.float().cpu()).numpy()
+        # deep-forest fallback is the strict-order general kernel: bitwise
+        np.testing.assert_array_equal(
+            s_gpu.view(np.int32), s_cpu.view(np.int32))
+
+    def test_standard_end_to_end_max_samples_8192(self, dev):
+        X = torch.from_numpy(make_data(30000, 10, seed=63)).to(dev)
+        model = IsolationForest(
+            numEstimators=16, maxSamples=8192.0, randomSeed=17).fit(X)
+        s_gpu = model.score(X).cpu().numpy()
+        s_cpu = model.score(X.float().cpu()).numpy()
+        np.testing.assert_array_equal(
+            s_gpu.view(np.int32), s_cpu.view(np.int32))

@@ -42,8 +42,8 @@
 #include <vector>
 
 namespace ifa {
-void launch_score_forest(bool bf16, int rpt, bool rows_lds, int ilp,
-                         const void* X, const void* nodes,
+void launch_score_forest(bool bf16, int rpt, bool rows_lds, bool nodes_lds,
+                         int ilp, const void* X, const void* nodes,
                          const int32_t* ncount, float* out, int64_t N,
                          int32_t d, int32_t dpad, int32_t Tpad,
                          int32_t max_nodes, int32_t height_limit, float fT,
@@ -608,9 +608,10 @@ int main(int argc, char** argv) {
     HIP_CHECK(hipMemcpy(dNcount, p.ncount.data(), p.ncount.size() * 4,
                         hipMemcpyHostToDevice));
     size_t node_bytes = (size_t)4 * p.mn * 8;
-    if (node_bytes > kMaxLds) {
-      fprintf(stderr, "error: tree too large for LDS staging\n");
-      return 2;
+    bool nodes_lds = true;
+    if (node_bytes + (size_t)256 * dpad * elem > 150 * 1024) {
+      nodes_lds = false;
+      node_bytes = 0;
     }
     int rpt = 2;
     bool rows_lds = true;
@@ -627,7 +628,7 @@ int main(int argc, char** argv) {
     int64_t rows_per_block = (int64_t)(rows_lds ? rpt : 1) * 256;
     int blocks = (int)std::min<int64_t>(
         (N + rows_per_block - 1) / rows_per_block, 8192);
-    ifa::launch_score_forest(bf16, rpt, rows_lds, 4, dX, dNodes,
+    ifa::launch_score_forest(bf16, rpt, rows_lds, nodes_lds, 4, dX, dNodes,
                              (const int32_t*)dNcount, (float*)dOut, N, d,
                              (int32_t)dpad, p.Tpad, p.mn, p.max_depth,
                              (float)p.T, c_norm, 1, lds, blocks, 0);
@@ -686,7 +687,9 @@ int main(int argc, char** argv) {
       T_report = p.T;
       size_t lds = (size_t)p.mn * 12 + 16 + (size_t)p.mn * (D / 4 + 1) * 16;
       if (lds > kMaxLds) {
-        fprintf(stderr, "error: tree too large for LDS staging\n");
+        fprintf(stderr,
+                "error: extended model too deep for the native dense "
+                "kernel; use the Python engine\n");
         return 2;
       }
       void *dNodes, *dVals, *dHw, *dNcount;
