@@ -101,16 +101,32 @@ def one_case(rs: np.random.RandomState, it: int) -> str:
             _gpu_forest_cache = {}
         gpu_ps = gpu_engine.score_extended_forest(Shell(), Xt, finalize=False)
         diff = np.abs(gpu_ps.cpu().numpy() - cpu_ps)
+        # mirror the routing to know the numeric contract: sparse v2 and the
+        # general kernel keep strict j-order (bitwise); dense reassociates
         nnz = min(ext_level + 1, k)
-        if nnz <= 5:  # sparse v2 or general: strict j-order, bitwise
-            if not np.array_equal(gpu_ps.cpu().numpy().view(np.int32),
-                                  cpu_ps.view(np.int32)):
-                return f"SCORE MISMATCH EIF sparse (bitwise): {desc}"
-        else:  # dense: reassociated dot, knife-edge tolerance
+        mn = gpu_f.feature.shape[1]
+        elem = 2 if bf16 else 4
+        dpad_s = d
+        while (dpad_s % 4 != 2) if bf16 else (dpad_s % 2 != 1):
+            dpad_s += 1
+        sparse_route = (nnz <= 5
+                        and mn * (12 + nnz * 8) + 2 * 256 * dpad_s * elem
+                        <= 150 * 1024)
+        D = 8 if d <= 8 else (16 if d <= 16 else 32)
+        dense_route = (not sparse_route and d <= 32
+                       and (nnz == d or nnz >= 6)
+                       and mn * 12 + 16 + mn * (D // 4 + 1) * 16 <= 160 * 1024)
+        old_dense_route = (not sparse_route and not dense_route
+                           and nnz == d and d % 4 == 0 and mn * 8 <= 120 * 1024)
+        if dense_route or old_dense_route:
             frac = float((diff > 1e-3 * max(1.0, np.abs(cpu_ps).max())).mean())
             if frac > 0.01:
                 return (f"SCORE MISMATCH EIF dense (>{frac:.4f} rows off): "
                         f"{desc}")
+        else:  # sparse v2 or general strict-order kernel: bitwise
+            if not np.array_equal(gpu_ps.cpu().numpy().view(np.int32),
+                                  cpu_ps.view(np.int32)):
+                return f"SCORE MISMATCH EIF strict-order (bitwise): {desc}"
     return ""
 
 
