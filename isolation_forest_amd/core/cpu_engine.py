@@ -23,7 +23,7 @@ SURVEY.md §7 "Hard parts" #1. Self-determinism is bitwise: the same seed
 yields the same forest on CPU and GPU, at any world size, because every
 draw is keyed by (seed, purpose, treeId, nodeId/slot, attempt).
 
-PRECISION CONTRACT (kept identical in ops/hip/iforest_kernels.hip):
+PRECISION CONTRACT (kept identical in ops/hip/forest_kernels.hip):
 * features are float32 (core/Utils.scala:11),
 * node min/max scans are float32,
 * the split point / hyperplane offset is computed in float64 and then
