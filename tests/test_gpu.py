@@ -493,3 +493,17 @@ class TestDeepForests:
         s_cpu = model.score(X.float().cpu()).numpy()
         np.testing.assert_array_equal(
             s_gpu.view(np.int32), s_cpu.view(np.int32))
+
+
+class TestParityFuzzSmoke:
+    def test_ten_random_configs(self, dev):
+        import subprocess
+        import sys
+        import os
+
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        out = subprocess.run(
+            [sys.executable, os.path.join(repo, "tools", "fuzz_parity.py"),
+             "--iters", "10", "--seed", "42"],
+            capture_output=True, text=True, timeout=600, cwd=repo)
+        assert out.returncode == 0, out.stdout + out.stderr

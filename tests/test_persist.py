@@ -388,3 +388,16 @@ class TestSnappyWrite:
         for n in [0, 1, 59, 60, 61, 255, 256, 65535, 65536, 1 << 20]:
             data = bytes((i * 7 + n) & 0xFF for i in range(n))
             assert snappy_decompress(snappy_compress_literal(data)) == data
+
+
+class TestPersistFuzzSmoke:
+    def test_five_random_models(self):
+        import subprocess
+        import sys
+
+        out = subprocess.run(
+            [sys.executable, os.path.join(
+                os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                "tools", "fuzz_persist.py"), "--iters", "5", "--seed", "42"],
+            capture_output=True, text=True, timeout=600)
+        assert out.returncode == 0, out.stdout + out.stderr
