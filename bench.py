@@ -16,11 +16,45 @@ with random seeds. Rank 0 prints ONE JSON line.
 import argparse
 import json
 import os
+import socket
 import sys
 import time
 
 import numpy as np
 import torch
+
+
+def _free_port() -> int:
+    s = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _self_launch(n: int) -> int:
+    """Re-exec this script under torch.distributed.run with n ranks.
+
+    The driver may call ``python bench.py --gpus N`` directly; the contract
+    requires one rank per GPU over RCCL, so when WORLD_SIZE is unset we
+    self-launch through the elastic launcher (127.0.0.1 rendezvous — the
+    container hostname may not resolve).
+    """
+    import subprocess
+
+    cmd = [
+        sys.executable,
+        "-m",
+        "torch.distributed.run",
+        "--nnodes=1",
+        f"--nproc-per-node={n}",
+        "--master-addr=127.0.0.1",
+        f"--master-port={_free_port()}",
+        os.path.abspath(__file__),
+    ] + sys.argv[1:]
+    env = dict(os.environ)
+    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    return subprocess.call(cmd, env=env)
 
 
 def make_data(rows, d, device, dtype, seed, outlier_frac=0.005):
@@ -76,6 +110,9 @@ def main():
     ap.add_argument("--contamination", type=float, default=0.0)
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
+
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        sys.exit(_self_launch(args.gpus))
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
     from isolation_forest_amd import ExtendedIsolationForest, IsolationForest
@@ -175,6 +212,7 @@ def main():
                 "rows_total": rows_total,
                 "features": args.features,
                 "parallelism": f"row-sharded dp{world_size} + tree-sharded build",
+                "world_size": world_size,
                 "contamination": args.contamination,
                 "auroc": round(measured_auroc, 4),
                 "device": device.split(":")[0],

@@ -75,6 +75,22 @@ class TestBenchContract:
         assert len(lines) == 1, out.stdout
         r = self._check_payload(lines[0], 2)
         assert r["n_gpus"] in (0, 2)  # 0 = CPU plumbing mode
+        assert r["config"]["world_size"] == 2
+
+    def test_direct_gpus_flag_self_launches(self):
+        """`python bench.py --gpus 2` invoked DIRECTLY must spawn 2 ranks
+        (VERDICT.md round-1 gap #2: the driver runs exactly this)."""
+        out = subprocess.run(
+            [sys.executable, "bench.py", "--gpus", "2", "--steps", "1",
+             "--warmup", "0", "--rows", "4000", "--trees", "16"],
+            cwd=REPO, capture_output=True, text=True, timeout=600,
+        )
+        assert out.returncode == 0, out.stderr[-2000:]
+        lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+        assert len(lines) == 1, out.stdout
+        r = self._check_payload(lines[0], 2)
+        assert r["config"]["world_size"] == 2
+        assert r["config"]["rows_total"] == 8000
 
 
 class TestTimingSync:
