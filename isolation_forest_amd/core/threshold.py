@@ -136,6 +136,14 @@ def compute_threshold(
         vals = sel.double()
     residual = k - rank_below_lo
     if residual <= 0:
+        # rank bookkeeping put the k-th score strictly below the bracket
+        # floor: re-select exactly over the elements below it so the result
+        # is an actual score value (exact order-statistic contract), not
+        # the float64 bin edge.
+        sel = s[s < lo32]
+        vals = comm.all_gather_1d(sel.double()) if comm is not None else sel.double()
+        if vals.numel() >= k:
+            return float(torch.kthvalue(vals, k).values.item())
         return float(lo_v)
     if residual > vals.numel():
         # bracket missed the target (pathological edge rounding): fall back

@@ -234,7 +234,8 @@ def _nodes_packed_v4(forest, d_sentinel: int, bf16: bool):
     """v4 packed nodes for score_forest_v4 (see forest_kernels.hip):
     w0 = feat | right<<12; internal w1 = integer key threshold; leaf:
     feat = d_sentinel, right = own id (self-loop), w1 = f32(depth + c(m)).
-    Trees are padded to a multiple of 4 with single-leaf value-0 dummies.
+    Trees are padded to a multiple of 8 with single-leaf value-0 dummies
+    (bindings.cpp enforces Tpad % 8 == 0).
     Returns (packed int32 [Tpad, mn, 2], ncount int32 [Tpad])."""
     T, mn = forest.feature.shape
     if mn > 32767:

@@ -75,9 +75,16 @@ class IsolationForestSKL:
     def fit(self, X, y=None):
         contamination = self.contamination
         auto = contamination == "auto"
+        X_t = self._to_tensor(X)
+        n_rows = int(X_t.shape[0])
         ms = self.max_samples
         if ms == "auto":
             ms = 256
+        if not isinstance(ms, float) or ms > 1.0:
+            # sklearn contract: integer/"auto" max_samples is clamped to
+            # n_samples (sklearn's IsolationForest uses min(256, n)); the
+            # engine would otherwise reject maxSamples > row count.
+            ms = min(int(ms), n_rows)
         est = _Engine(
             numEstimators=int(self.n_estimators),
             maxSamples=float(ms),
@@ -88,7 +95,7 @@ class IsolationForestSKL:
             randomSeed=int(self.random_state)
             if self.random_state is not None else 1,
         )
-        self.model_ = est.fit(self._to_tensor(X))
+        self.model_ = est.fit(X_t)
         if auto:
             # conventional score cut at 0.5 (Liu et al.: scores above 0.5
             # indicate anomalies)

@@ -276,6 +276,28 @@ class TestSklearnAdapter:
         labels = clf.predict(X)
         assert (labels == -1).any() or (labels == 1).all()
 
+    def test_small_dataset_clamps_max_samples(self):
+        """default max_samples=256 must clamp to n_samples for
+        small datasets (sklearn's min(256, n) contract)."""
+        from isolation_forest_amd.sklearn_api import IsolationForestSKL
+
+        rs = np.random.RandomState(7)
+        X = rs.normal(size=(50, 4)).astype(np.float32)
+        clf = IsolationForestSKL(n_estimators=20, random_state=1)
+        labels = clf.fit(X).predict(X)  # must not raise
+        assert labels.shape == (50,)
+        assert clf.model_.num_samples == 50
+        # explicit integer max_samples larger than n also clamps
+        clf2 = IsolationForestSKL(n_estimators=10, max_samples=1000,
+                                  random_state=1)
+        clf2.fit(X)
+        assert clf2.model_.num_samples == 50
+        # fractional max_samples unaffected
+        clf3 = IsolationForestSKL(n_estimators=10, max_samples=0.5,
+                                  random_state=1)
+        clf3.fit(X)
+        assert clf3.model_.num_samples == 25
+
     def test_sklearn_pipeline_compatible(self, gaussian_data):
         from sklearn.pipeline import Pipeline
         from sklearn.preprocessing import StandardScaler
