@@ -7,6 +7,8 @@ import pytest
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 REFERENCE_RESOURCES = "/root/reference/isolation-forest/src/test/resources"
+FIXTURES = os.path.join(os.path.dirname(os.path.abspath(__file__)), "fixtures")
+GOLDEN = os.path.join(FIXTURES, "golden")
 
 
 def pytest_configure(config):
@@ -72,15 +74,22 @@ def auroc(y_true, scores):
     return (ranks[pos].sum() - n_pos * (n_pos + 1) / 2.0) / (n_pos * n_neg)
 
 
+def _load_odds(name):
+    """In-repo ODDS fixture (committed f32 npz, VERDICT r01 Missing #4) —
+    features were float32 in the reference engine anyway (Utils.scala:11)."""
+    path = os.path.join(FIXTURES, f"{name}.npz")
+    with np.load(path) as z:
+        return z["X"].astype(np.float32), z["y"].astype(np.float64)
+
+
 @pytest.fixture(scope="session")
 def mammography():
-    """The ODDS mammography dataset from the reference checkout, when present
-    (11183 rows x 6 features + label). Skips cleanly when unavailable
-    (e.g. on a GPU box where only /root/repo is snapshotted)."""
-    path = os.path.join(REFERENCE_RESOURCES, "mammography.csv")
-    if not os.path.exists(path):
-        pytest.skip("reference mammography.csv not available")
-    raw = np.loadtxt(path, delimiter=",", dtype=np.float64)
-    X = raw[:, :-1].astype(np.float32)
-    y = raw[:, -1]
-    return X, y
+    """The ODDS mammography dataset (11183 rows x 6 features + label),
+    committed in-repo so quality gates run on GPU boxes too."""
+    return _load_odds("mammography")
+
+
+@pytest.fixture(scope="session")
+def shuttle():
+    """The ODDS shuttle dataset (49097 rows x 9 features + label)."""
+    return _load_odds("shuttle")

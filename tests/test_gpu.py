@@ -232,6 +232,43 @@ class TestEndToEndGPU:
         np.testing.assert_allclose(s_cpu, s_gpu, rtol=0, atol=2e-7)
 
 
+class TestRealDataQualityGPU:
+    """Real-data AUROC gates on the GPU path (VERDICT r01 Missing #4):
+    in-repo ODDS fixtures make these run on the driver's GPU box, where
+    synthetic-only gates (bench AUROC 1.0) are non-discriminating.
+    Reference bars: IsolationForestTest.scala:77-88 (mammography 0.86),
+    :229-236 (shuttle > 0.99)."""
+
+    def test_mammography_auroc_gpu(self, dev, mammography):
+        X, y = mammography
+        Xt = torch.from_numpy(X).to(dev)
+        model = IsolationForest(
+            numEstimators=100, maxSamples=256.0, randomSeed=1
+        ).fit(Xt)
+        scores = model.score(Xt).cpu().numpy()
+        assert auroc(y, scores) == pytest.approx(0.86, abs=0.02)
+
+    def test_mammography_auroc_gpu_extended(self, dev, mammography):
+        X, y = mammography
+        Xt = torch.from_numpy(X).to(dev)
+        model = ExtendedIsolationForest(
+            numEstimators=100, maxSamples=256.0, randomSeed=1
+        ).fit(Xt)
+        scores = model.score(Xt).cpu().numpy()
+        assert auroc(y, scores) == pytest.approx(0.86, abs=0.02)
+
+    def test_shuttle_auroc_gpu(self, dev, shuttle):
+        X, y = shuttle
+        Xt = torch.from_numpy(X).to(dev)
+        model = IsolationForest(
+            numEstimators=100, maxSamples=256.0, randomSeed=1
+        ).fit(Xt)
+        scores = model.score(Xt).cpu().numpy()
+        assert auroc(y, scores) > 0.99
+        assert float(scores[y == 1].mean()) == pytest.approx(0.61, abs=0.02)
+        assert float(scores[y == 0].mean()) == pytest.approx(0.41, abs=0.02)
+
+
 class TestDevicePacking:
     """The torch-side scoring pack (built from raw device build outputs)
     must be BITWISE identical to the host numpy pack."""

@@ -203,6 +203,31 @@ class TestMammographyQuality:
         assert auroc(y, score) == pytest.approx(0.86, abs=0.02)
 
 
+class TestShuttleQuality:
+    """The reference's shuttle gates (IsolationForestTest.scala:170-239):
+    AUROC > 0.99 and labeled outlier/inlier mean scores 0.61/0.41 +/- 0.02."""
+
+    def test_standard_auroc_and_score_means(self, shuttle):
+        X, y = shuttle
+        model = IsolationForest(
+            numEstimators=100, maxSamples=256.0, randomSeed=1
+        ).fit(X)
+        score = model.transform(X)["outlierScore"].numpy()
+        assert auroc(y, score) > 0.99
+        assert float(score[y == 1].mean()) == pytest.approx(0.61, abs=0.02)
+        assert float(score[y == 0].mean()) == pytest.approx(0.41, abs=0.02)
+
+    def test_extended_auroc(self, shuttle):
+        """ExtendedIF on shuttle > 0.99 (ExtendedIsolationForestTest:333-373,
+        ext=8 = fully extended for d=9)."""
+        X, y = shuttle
+        model = ExtendedIsolationForest(
+            numEstimators=100, maxSamples=256.0, randomSeed=1
+        ).fit(X)
+        score = model.transform(X)["outlierScore"].numpy()
+        assert auroc(y, score) > 0.99
+
+
 class TestExactQuantileStress:
     """compute_threshold exact mode must return the true order statistic
     even on tie-heavy and tightly-clustered score distributions (the

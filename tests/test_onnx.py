@@ -200,36 +200,28 @@ class TestWireFormat:
             assert k in meta["paramMap"]
 
 
-REFERENCE_RESOURCES = "/root/reference/isolation-forest/src/test/resources"
-needs_reference = pytest.mark.skipif(
-    not os.path.isdir(REFERENCE_RESOURCES),
-    reason="reference checkout not available",
-)
+from tests.conftest import GOLDEN
 
 
-@needs_reference
 class TestReferenceModelConversion:
     """Convert the REFERENCE's golden Spark-written model through our
     converter (the reference converter's exact file contract) and check
     score parity against our engine loading the same model — the analog of
     the reference's two-part Spark<->ONNX integration test at 1e-5."""
 
-    def test_reference_golden_model_onnx_parity(self):
+    def test_reference_golden_model_onnx_parity(self, mammography):
         import glob as _glob
 
         from isolation_forest_amd import IsolationForestModel
 
-        mdir = os.path.join(REFERENCE_RESOURCES, "savedIsolationForestModel")
+        mdir = os.path.join(GOLDEN, "savedIsolationForestModel")
         avro_file = _glob.glob(os.path.join(mdir, "data", "*.avro"))[0]
         meta_file = _glob.glob(os.path.join(mdir, "metadata", "part-*"))[0]
         conv = IsolationForestConverter(avro_file, meta_file)
         onnx_bytes = conv.convert().serialize()
 
         model = IsolationForestModel.load(mdir)
-        data = np.loadtxt(
-            os.path.join(REFERENCE_RESOURCES, "mammography.csv"),
-            delimiter=",", comments="#")
-        X = data[:2000, :-1].astype(np.float32)
+        X = mammography[0][:2000]
         engine_scores = model.score(torch.from_numpy(X)).numpy()
         onnx_scores = evaluator.run(onnx_bytes, X)["outlier_score"].ravel()
         assert np.abs(onnx_scores - engine_scores).max() < 1e-5

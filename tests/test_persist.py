@@ -19,7 +19,7 @@ from isolation_forest_amd import (
     IsolationForestModel,
 )
 from isolation_forest_amd.persist import avro_io, model_io
-from tests.conftest import REFERENCE_RESOURCES, auroc
+from tests.conftest import GOLDEN, auroc
 
 
 @pytest.fixture(scope="module")
@@ -166,42 +166,37 @@ class TestModelRoundTrip:
         assert loaded.getContamination() == 0.1
 
 
-needs_reference = pytest.mark.skipif(
-    not os.path.isdir(REFERENCE_RESOURCES),
-    reason="reference checkout not available",
-)
-
-
-@needs_reference
 class TestReferenceGoldenFixtures:
-    """Load the models committed to the reference repo (written by Spark,
-    snappy-coded Avro) and reproduce their golden structure exactly —
-    the backward-compat anchor (IsolationForestModelWriteReadTest:391-408)."""
+    """Load the golden Spark-written models (snappy-coded Avro, committed
+    in-repo under tests/fixtures/golden — byte-identical copies of the
+    reference's 2018 fixtures) and reproduce their golden structure
+    exactly — the backward-compat anchor
+    (IsolationForestModelWriteReadTest:391-408)."""
 
     def test_load_standard_golden(self):
-        path = os.path.join(REFERENCE_RESOURCES, "savedIsolationForestModel")
+        path = os.path.join(GOLDEN, "savedIsolationForestModel")
         model = IsolationForestModel.load(path)
         assert model.forest.num_trees == 100
         assert model.num_samples == 256
         assert model.num_features == 6
         assert model.outlier_score_threshold == pytest.approx(0.6015323679815825)
         # the golden file is tree 0's toString (savedIsolationForestModelTreeStructureTest)
-        golden = open(os.path.join(REFERENCE_RESOURCES, "expectedTreeStructure.txt")).read()
+        golden = open(os.path.join(GOLDEN, "expectedTreeStructure.txt")).read()
         assert model.forest.tree_to_string(0) == golden
 
     def test_load_extended_golden(self):
-        path = os.path.join(REFERENCE_RESOURCES, "savedExtendedIsolationForestModel")
+        path = os.path.join(GOLDEN, "savedExtendedIsolationForestModel")
         model = ExtendedIsolationForestModel.load(path)
         assert model.forest.num_trees == 100
         golden = open(
-            os.path.join(REFERENCE_RESOURCES, "expectedExtendedTreeStructure.txt")
+            os.path.join(GOLDEN, "expectedExtendedTreeStructure.txt")
         ).read()
         assert model.forest.tree_to_string(0) == golden
 
     def test_golden_model_scores_mammography(self, mammography):
         """Our scorer over Spark's 2018 model reproduces the reference AUROC."""
         X, y = mammography
-        path = os.path.join(REFERENCE_RESOURCES, "savedIsolationForestModel")
+        path = os.path.join(GOLDEN, "savedIsolationForestModel")
         model = IsolationForestModel.load(path)
         scores = model.score(torch.from_numpy(X)).numpy()
         assert auroc(y, scores) == pytest.approx(0.86, abs=0.02)
@@ -211,7 +206,7 @@ class TestReferenceGoldenFixtures:
 
     def test_roundtrip_reference_model(self, tmp_path):
         """save(load(reference)) preserves every node record exactly."""
-        src = os.path.join(REFERENCE_RESOURCES, "savedIsolationForestModel")
+        src = os.path.join(GOLDEN, "savedIsolationForestModel")
         model = IsolationForestModel.load(src)
         dst = str(tmp_path / "rt")
         model.save(dst)
