@@ -18,6 +18,7 @@ dense v2 (tolerance), else the general strict-order kernel.
 from __future__ import annotations
 
 import math
+import os
 
 import numpy as np
 import torch
@@ -509,7 +510,8 @@ def _device_forest(model, device, v4_key=None):
                 extra["hw"] = torch.from_numpy(
                     np.ascontiguousarray(forest.hyper_w)).to(device)
             extra["height"] = max_depth
-        elif isinstance(v4_key, tuple) and v4_key[0] == "eif_dense":
+        elif isinstance(v4_key, tuple) and v4_key[0] in ("eif_dense",
+                                                         "eif_dense3"):
             D = v4_key[1]
             if raw is not None:
                 aos, values_t, hw_t, max_depth = _eif_dense_packed_device(
@@ -526,6 +528,15 @@ def _device_forest(model, device, v4_key=None):
                 ).to(device)
                 extra["values"] = torch.from_numpy(values).to(device)
                 extra["hw"] = torch.from_numpy(hw_dense).to(device)
+            if v4_key[0] == "eif_dense3":
+                # v3 weight staging: RNE-round the trained f32 weights to
+                # bf16 and pack pairs (j even low, j+1 high) into int32 for
+                # the v_dot2c_f32_bf16 dot (PARITY.md EIF bf16 contract)
+                hb = extra["hw"].to(torch.bfloat16).view(torch.uint16)
+                hb = hb.to(torch.int32)
+                extra["hwp"] = (
+                    hb[..., 0::2] | (hb[..., 1::2] << 16)
+                ).contiguous()
             extra["height"] = max_depth
         elif v4_key is not None:
             d, bf16 = v4_key
@@ -549,7 +560,10 @@ def _device_forest(model, device, v4_key=None):
                 extra["hw"] = torch.from_numpy(
                     np.ascontiguousarray(forest.hyper_w)
                 ).to(device)
-        cache.clear()  # one cached device copy per model is enough
+        # bounded FIFO (a few packings per model: e.g. alternating bf16/f32
+        # scoring of one model keeps both — VERDICT r01 weak #6)
+        while len(cache) >= 4:
+            cache.pop(next(iter(cache)))
         cache[key] = (aos, ncount, extra)
     return cache[key]
 
@@ -612,6 +626,17 @@ def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torc
     if d <= 32 and (nnz == d or nnz >= 6):
         D = 8 if d <= 8 else (16 if d <= 16 else 32)
         mn = forest.feature.shape[1]
+        use_v3 = (X.dtype == torch.bfloat16
+                  and os.environ.get("IFA_EIF_DENSE_V2") != "1")
+        if use_v3:
+            lds = mn * 12 + 16 + mn * (D // 8 + 1) * 16
+            if lds <= 160 * 1024:
+                aos, ncount, extra = _device_forest(
+                    model, X.device, v4_key=("eif_dense3", D))
+                return ext.score_extended_dense_v3(
+                    X.contiguous(), aos, extra["values"], extra["hwp"],
+                    ncount, extra["height"], c, finalize,
+                )
         lds = mn * 12 + 16 + mn * (D // 4 + 1) * 16
         if lds <= 160 * 1024:
             aos, ncount, extra = _device_forest(

@@ -77,6 +77,15 @@ void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
                                     float c_norm, int finalize, size_t lds,
                                     int blocks, hipStream_t stream);
 
+void launch_score_extended_dense_v3(int D, const void* X, const void* nodes,
+                                    const float* values, const uint32_t* hwp,
+                                    const int32_t* ncount, float* out,
+                                    int64_t N, int32_t d, int32_t T,
+                                    int32_t max_nodes, int32_t height_limit,
+                                    float fT, float c_norm, int finalize,
+                                    size_t lds, int blocks,
+                                    hipStream_t stream);
+
 }  // namespace ifa
 
 namespace {
@@ -386,6 +395,48 @@ torch::Tensor score_extended_dense_v2(torch::Tensor X,
   return out;
 }
 
+torch::Tensor score_extended_dense_v3(torch::Tensor X,
+                                      torch::Tensor nodes_packed,
+                                      torch::Tensor values, torch::Tensor hwp,
+                                      torch::Tensor ncount,
+                                      int64_t height_limit, double c_norm,
+                                      bool finalize) {
+  CHECK_CUDA(X);
+  CHECK_CONTIG(X);
+  CHECK_CUDA(nodes_packed);
+  CHECK_CONTIG(nodes_packed);
+  CHECK_CUDA(values);
+  CHECK_CONTIG(values);
+  CHECK_CUDA(hwp);
+  CHECK_CONTIG(hwp);
+  CHECK_CUDA(ncount);
+  check_x(X);
+  TORCH_CHECK(is_bf16(X), "dense v3 requires bf16 rows (f32 rows route to v2)");
+  TORCH_CHECK(nodes_packed.dim() == 3 && nodes_packed.size(2) == 2 &&
+                  nodes_packed.scalar_type() == torch::kInt32,
+              "nodes must be packed int32 [T, max_nodes, 2]");
+  TORCH_CHECK(hwp.dim() == 3 && hwp.scalar_type() == torch::kInt32,
+              "hwp must be int32 [T, max_nodes, D/2] packed bf16 pairs");
+  int64_t N = X.size(0), d = X.size(1);
+  int64_t T = nodes_packed.size(0), max_nodes = nodes_packed.size(1);
+  TORCH_CHECK(d <= 32, "dense v3 supports d <= 32");
+  const int D = d <= 8 ? 8 : (d <= 16 ? 16 : 32);
+  TORCH_CHECK(hwp.size(2) == D / 2, "hwp must be packed to D/2 dwords");
+  auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
+  if (N == 0) return out;
+
+  size_t lds = (size_t)max_nodes * 12 + 16 + (size_t)max_nodes * (D / 8 + 1) * 16;
+  TORCH_CHECK(lds <= kMaxLds, "tree too large for LDS staging");
+  int blocks = (int)std::min<int64_t>((N + 511) / 512, 8192);
+  ifa::launch_score_extended_dense_v3(
+      D, X.data_ptr(), nodes_packed.data_ptr<int32_t>(),
+      values.data_ptr<float>(), (const uint32_t*)hwp.data_ptr<int32_t>(),
+      ncount.data_ptr<int32_t>(), out.data_ptr<float>(), N, (int32_t)d,
+      (int32_t)T, (int32_t)max_nodes, (int32_t)height_limit, (float)T,
+      (float)c_norm, finalize ? 1 : 0, lds, blocks, current_stream());
+  return out;
+}
+
 torch::Tensor score_extended_sparse_v2(torch::Tensor X,
                                        torch::Tensor nodes_packed,
                                        torch::Tensor values,
@@ -446,5 +497,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "sparse EIF scoring, fixed-trip batched walk (K7 small-nnz path)");
   m.def("score_extended_dense_v2", &score_extended_dense_v2,
         "dense EIF scoring, rows-in-registers (K7 fast path)");
+  m.def("score_extended_dense_v3", &score_extended_dense_v3,
+        "dense EIF scoring, packed-bf16 weights + v_dot2c (K7 bf16 path)");
   m.attr("WAVE") = 64;
 }

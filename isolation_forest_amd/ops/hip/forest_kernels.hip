@@ -943,6 +943,151 @@ __global__ void __launch_bounds__(EIFD_THREADS, 4) score_extended_dense_v2(
 }
 
 // ---------------------------------------------------------------------------
+// extended scoring, dense fast path v3 — bf16 INPUT rows only.
+//
+// The v2 kernel is bound by its per-visit 128-B f32 LDS weight-row read
+// (PMC: 27% LDS conflict cycles, WAIT_ANY 44% — profiles/
+// r01_bench_and_kernels.md). v3 halves that traffic: weights are staged in
+// LDS as PACKED bf16 (round-to-nearest-even from the trained f32 weights,
+// done once on device), rows are kept as packed-bf16 register pairs (the
+// input is already bf16, so the pack is exact), and the dot runs on
+// v_dot2c_f32_bf16 (2 MACs/instr, f32 accumulate) — 4 partial accumulator
+// chains, same reassociation family as v2. Per wave-visit: 4x ds_read_b128
+// (16 LDS-array cycles before conflicts) instead of v2's 8 (32 cycles),
+// and ~half the dot VALU. Numerics: weights are bf16-rounded => knife-edge
+// walk decisions can differ from the f32-weight oracle; the contract is
+// documented in PARITY.md and tested against a bf16-weight oracle with the
+// same isolated-flip allowance as the v2 dense route. f32 input rows keep
+// the v2 kernel (unchanged f32-weight numerics).
+// ---------------------------------------------------------------------------
+
+typedef __bf16 bf16x2_t __attribute__((ext_vector_type(2)));
+
+__device__ __forceinline__ float dot2_bf16(uint32_t w, uint32_t x, float acc) {
+  union { uint32_t u; bf16x2_t v; } cw, cx;
+  cw.u = w;
+  cx.u = x;
+  return __builtin_amdgcn_fdot2_f32_bf16(cw.v, cx.v, acc, false);
+}
+
+template <int D, int RPT>
+__global__ void __launch_bounds__(EIFD_THREADS, 6) score_extended_dense_v3(
+    const uint16_t* __restrict__ X,     // raw bf16 bits [N][d]
+    const int2* __restrict__ nodes,     // [T][max_nodes] {right<<12, offset/-inf}
+    const float* __restrict__ values,   // [T][max_nodes] leaf value+depth
+    const uint32_t* __restrict__ hwp,   // [T][max_nodes][D/2] packed bf16 pairs
+    const int32_t* __restrict__ ncnt,   // [T]
+    float* __restrict__ out, int64_t N, int32_t d, int32_t T,
+    int32_t max_nodes, int32_t height_limit, float fT, float c_norm,
+    int32_t finalize) {
+  const int tid = threadIdx.x;
+  const int rows_per_iter = RPT * EIFD_THREADS;
+  constexpr int PW4 = D / 8 + 1;  // LDS weight-row stride in uint4s
+
+  int2* tlds = (int2*)smem;                  // [max_nodes]
+  float* vlds = (float*)(tlds + max_nodes);  // [max_nodes]
+  uint4* wlds = (uint4*)(((uintptr_t)(vlds + max_nodes) + 15) & ~15ull);
+
+  for (int64_t block_row0 = (int64_t)blockIdx.x * rows_per_iter; block_row0 < N;
+       block_row0 += (int64_t)gridDim.x * rows_per_iter) {
+    uint32_t xp[RPT][D / 2];  // packed bf16 feature pairs, exact
+#pragma unroll
+    for (int r = 0; r < RPT; ++r) {
+      const int64_t my_row = block_row0 + tid + r * EIFD_THREADS;
+      const bool ok = my_row < N;
+      const int64_t base = my_row * d;
+#pragma unroll
+      for (int j2 = 0; j2 < D / 2; ++j2) {
+        const uint32_t lo = (ok && 2 * j2 < d) ? X[base + 2 * j2] : 0u;
+        const uint32_t hi = (ok && 2 * j2 + 1 < d) ? X[base + 2 * j2 + 1] : 0u;
+        xp[r][j2] = lo | (hi << 16);
+      }
+    }
+
+    float psum[RPT];
+#pragma unroll
+    for (int r = 0; r < RPT; ++r) psum[r] = 0.f;
+
+    for (int t = 0; t < T; ++t) {
+      __syncthreads();
+      const int nc = ncnt[t];
+      {
+        const int2* ss = nodes + (int64_t)t * max_nodes;
+        const float* vs = values + (int64_t)t * max_nodes;
+        for (int i = tid; i < nc; i += EIFD_THREADS) {
+          tlds[i] = ss[i];
+          vlds[i] = vs[i];
+        }
+        // packed rows are D/8 uint4s wide in global, PW4 in LDS
+        const uint4* ws = (const uint4*)(hwp + (int64_t)t * max_nodes * (D / 2));
+        const int total4 = nc * (D / 8);
+        for (int g = tid; g < total4; g += EIFD_THREADS) {
+          const int i = g / (D / 8), j4 = g % (D / 8);
+          wlds[i * PW4 + j4] = ws[g];
+        }
+      }
+      __syncthreads();
+
+      int cur[RPT];
+#pragma unroll
+      for (int r = 0; r < RPT; ++r) cur[r] = 0;
+
+      for (int it = 0; it < height_limit; ++it) {
+        int2 nd[RPT];
+#pragma unroll
+        for (int r = 0; r < RPT; ++r) nd[r] = tlds[cur[r]];
+#pragma unroll
+        for (int r = 0; r < RPT; ++r) {
+          const int right = pn_right(nd[r].x);
+          if (right != cur[r]) {  // exec-mask finished walks out of the dot
+            // chunked reads bound live w registers (D=32 stays under the
+            // 80-VGPR budget that keeps 6 waves/SIMD resident)
+            constexpr int CHUNKS = (D >= 32) ? 2 : 1;
+            constexpr int CW4 = D / (8 * CHUNKS);  // uint4s per chunk
+            const uint4* wp = wlds + cur[r] * PW4;
+            float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+#pragma unroll
+            for (int ch = 0; ch < CHUNKS; ++ch) {
+              uint4 w[CW4];
+#pragma unroll
+              for (int j4 = 0; j4 < CW4; ++j4) w[j4] = wp[ch * CW4 + j4];
+#pragma unroll
+              for (int j4 = 0; j4 < CW4; ++j4) {
+                const int p = (ch * CW4 + j4) * 4;
+                a0 = dot2_bf16(w[j4].x, xp[r][p + 0], a0);
+                a1 = dot2_bf16(w[j4].y, xp[r][p + 1], a1);
+                a2 = dot2_bf16(w[j4].z, xp[r][p + 2], a2);
+                a3 = dot2_bf16(w[j4].w, xp[r][p + 3], a3);
+              }
+            }
+            const float dot = __fadd_rn(__fadd_rn(a0, a1),
+                                        __fadd_rn(a2, a3));
+            cur[r] = (dot < __int_as_float(nd[r].y)) ? cur[r] + 1 : right;
+          }
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < RPT; ++r)
+        psum[r] = __fadd_rn(psum[r], vlds[cur[r]]);
+    }
+
+#pragma unroll
+    for (int r = 0; r < RPT; ++r) {
+      const int64_t my_row = block_row0 + tid + r * EIFD_THREADS;
+      if (my_row < N) {
+        if (finalize) {
+          const float mean32 = (float)((double)psum[r] / (double)fT);
+          const double ratio = (double)mean32 / (double)c_norm;
+          out[my_row] = (float)exp2(-ratio);
+        } else {
+          out[my_row] = psum[r];
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // extended scoring, sparse fast path v2 (nnz <= 5, templated):
 // same node/value packing as the dense v2 kernel (self-looping leaves with
 // offset -inf and zero weights => fixed-trip walks), rows staged native in
@@ -1315,6 +1460,28 @@ void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
     else LSD2(uint32_t, 32, 1);
   }
 #undef LSD2
+}
+
+void launch_score_extended_dense_v3(int D, const void* X, const void* nodes,
+                                    const float* values, const uint32_t* hwp,
+                                    const int32_t* ncount, float* out,
+                                    int64_t N, int32_t d, int32_t T,
+                                    int32_t max_nodes, int32_t height_limit,
+                                    float fT, float c_norm, int finalize,
+                                    size_t lds, int blocks,
+                                    hipStream_t stream) {
+#define LSD3(DD, RR)                                                          \
+  do {                                                                        \
+    raise_lds((const void*)score_extended_dense_v3<DD, RR>, lds);             \
+    hipLaunchKernelGGL((score_extended_dense_v3<DD, RR>), dim3(blocks),       \
+                       dim3(EIFD_THREADS), lds, stream, (const uint16_t*)X,   \
+                       (const int2*)nodes, values, hwp, ncount, out, N, d, T, \
+                       max_nodes, height_limit, fT, c_norm, finalize);        \
+  } while (0)
+  if (D == 8) LSD3(8, 2);
+  else if (D == 16) LSD3(16, 2);
+  else LSD3(32, 1);
+#undef LSD3
 }
 
 void launch_score_extended_sparse_v2(bool bf16, int nnz, const void* X,

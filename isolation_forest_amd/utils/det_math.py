@@ -78,6 +78,17 @@ def det_cos2pi(u):
     return out
 
 
+def bf16_round(a: np.ndarray) -> np.ndarray:
+    """Round float32 values to bf16 (round-to-nearest-even), returned as
+    float32 — numpy mirror of torch's .to(torch.bfloat16) and of the GPU
+    dense-v3 weight staging (ops/hip/forest_kernels.hip). Finite inputs
+    only (hyperplane weights are finite by construction)."""
+    b = np.ascontiguousarray(a, dtype=np.float32).view(np.uint32)
+    r = ((b >> np.uint32(16)) & np.uint32(1)) + np.uint32(0x7FFF)
+    return ((b + r) & np.uint32(0xFFFF0000)).view(np.float32).reshape(
+        np.asarray(a).shape)
+
+
 def det_gaussian(u1, u2):
     """Box-Muller with deterministic log/cos; returns float64."""
     u1 = np.asarray(u1, dtype=np.float64)

@@ -157,6 +157,72 @@ class TestScoringParity:
         assert (diff > 1e-3).sum() <= 3
         assert np.quantile(diff, 0.999) < 1e-3
 
+    def test_extended_dense_v3_bf16_path_close(self, dev):
+        """bf16 rows route to dense v3 (packed-bf16 weights + v_dot2c).
+        Oracle: strict-order CPU walk with the SAME RNE-bf16-rounded
+        weights — remaining diffs are reassociation-level flips only."""
+        import copy
+
+        from isolation_forest_amd.ops import gpu_engine
+        from isolation_forest_amd.utils.det_math import bf16_round
+
+        rs = np.random.RandomState(11)
+        Xb = torch.from_numpy(
+            rs.normal(size=(3000, 8)).astype(np.float32)
+        ).to(torch.bfloat16)
+        X = Xb.float().numpy()  # exact f32 of the bf16 rows
+        bag = cpu_engine.sample_bags(3000, 16, 128, seed=5, bootstrap=False)
+        fs = cpu_engine.feature_subsets(8, 8, 16, seed=5)
+        forest = cpu_engine.build_extended_forest(X, bag, fs, 5, 128, 8, 8, 7)
+        f_b = copy.copy(forest)
+        f_b.hyper_w = bf16_round(forest.hyper_w)
+        cpu_ps = cpu_engine.path_lengths_extended(f_b, X)
+        model = ExtendedIsolationForest(numEstimators=16).fit(X[:600])
+        model.forest = forest
+        model._gpu_forest_cache = {}
+        gpu_ps = gpu_engine.score_extended_forest(
+            model, Xb.to(dev), finalize=False
+        )
+        diff = np.abs(gpu_ps.cpu().numpy() - cpu_ps)
+        assert (diff > 1e-3).sum() <= 5
+        assert np.quantile(diff, 0.99) < 1e-3
+
+    def test_extended_dense_v3_matches_v2_statistically(self, dev):
+        """v3 (bf16 weights) vs v2 (f32 weights) on the same bf16 rows:
+        path sums agree except for knife-edge walk flips; scores stay
+        rank-consistent (the documented bf16-weight precision cost)."""
+        import os as _os
+
+        from isolation_forest_amd.ops import gpu_engine
+
+        rs = np.random.RandomState(12)
+        Xb = torch.from_numpy(
+            rs.normal(size=(20000, 32)).astype(np.float32)
+        ).to(torch.bfloat16)
+        X32 = Xb.float().numpy()
+        model = ExtendedIsolationForest(numEstimators=64, randomSeed=3).fit(
+            X32
+        )
+        model._gpu_forest_cache = {}
+        s3 = gpu_engine.score_extended_forest(
+            model, Xb.to(dev), finalize=True
+        ).cpu().numpy()
+        _os.environ["IFA_EIF_DENSE_V2"] = "1"
+        try:
+            model._gpu_forest_cache = {}
+            s2 = gpu_engine.score_extended_forest(
+                model, Xb.to(dev), finalize=True
+            ).cpu().numpy()
+        finally:
+            del _os.environ["IFA_EIF_DENSE_V2"]
+        # scores: small perturbation, high rank agreement
+        assert np.abs(s3 - s2).max() < 0.05
+        assert np.abs(s3 - s2).mean() < 0.005
+        r2 = np.argsort(np.argsort(s2))
+        r3 = np.argsort(np.argsort(s3))
+        corr = np.corrcoef(r2, r3)[0, 1]
+        assert corr > 0.999
+
     def test_extended_path_sums_bitwise(self, dev):
         from isolation_forest_amd.ops import gpu_engine
 

@@ -83,3 +83,29 @@ class TestPhilox:
         assert r.min() >= 0 and r.max() <= 6
         counts = np.bincount(r, minlength=7)
         assert counts.min() > 0.9 * 100000 / 7
+
+
+class TestBf16Round:
+    def test_matches_torch_rne(self):
+        import torch
+
+        from isolation_forest_amd.utils.det_math import bf16_round
+
+        rs = np.random.RandomState(0)
+        a = rs.normal(size=50000).astype(np.float32)
+        a *= rs.choice([1e-8, 1.0, 1e8], 50000).astype(np.float32)
+        ours = bf16_round(a)
+        theirs = torch.from_numpy(a).to(torch.bfloat16).float().numpy()
+        np.testing.assert_array_equal(
+            ours.view(np.int32), theirs.view(np.int32)
+        )
+
+    def test_ties_to_even_and_shape(self):
+        from isolation_forest_amd.utils.det_math import bf16_round
+
+        # 1.0 + 2^-9 is exactly halfway between bf16 neighbours 1.0 and
+        # 1.0078125; RNE picks the even mantissa (1.0)
+        tie = np.float32(1.0 + 2.0 ** -9)
+        assert bf16_round(np.array([tie]))[0] == np.float32(1.0)
+        out = bf16_round(np.zeros((3, 4), dtype=np.float32))
+        assert out.shape == (3, 4)

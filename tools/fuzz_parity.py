@@ -119,6 +119,18 @@ def one_case(rs: np.random.RandomState, it: int) -> str:
         old_dense_route = (not sparse_route and not dense_route
                            and nnz == d and d % 4 == 0 and mn * 8 <= 120 * 1024)
         if dense_route or old_dense_route:
+            if bf16 and dense_route:
+                # v3 route: weights staged as bf16 (RNE) — oracle must use
+                # the same rounded weights; remaining diffs are
+                # reassociation-level only (PARITY.md EIF bf16 contract)
+                from isolation_forest_amd.utils.det_math import bf16_round
+
+                import copy as _copy
+                f_b = _copy.copy(cpu_f)
+                f_b.hyper_w = bf16_round(cpu_f.hyper_w)
+                cpu_ps_b = cpu_engine.path_lengths_extended(f_b, X)
+                diff = np.abs(gpu_ps.cpu().numpy() - cpu_ps_b)
+                cpu_ps = cpu_ps_b
             frac = float((diff > 1e-3 * max(1.0, np.abs(cpu_ps).max())).mean())
             if frac > 0.01:
                 return (f"SCORE MISMATCH EIF dense (>{frac:.4f} rows off): "
