@@ -34,12 +34,20 @@ class ExtendedIsolationForest(IsolationForest):
         feat_sub = cpu_engine.feature_subsets(
             rp.total_features, rp.num_features, T_local, seed, tree_id_offset=t_lo
         )
-        if X.is_cuda:
+        if X.is_cuda and rp.num_samples <= 16384:
             from ..ops import gpu_engine
 
             return gpu_engine.build_extended_forest(
                 X, bag_idx, feat_sub, seed, rp, tree_id_offset=t_lo
             )
+        if X.is_cuda:
+            import logging
+
+            logging.getLogger(__name__).warning(
+                "maxSamples %d exceeds the GPU build cap (16384); building "
+                "trees on CPU (scoring stays on GPU)", rp.num_samples,
+            )
+            X = X.float().cpu()
         Xc = X.contiguous().float().numpy()
         return cpu_engine.build_extended_forest(
             Xc, bag_idx, feat_sub, seed, rp.num_samples, rp.num_features,

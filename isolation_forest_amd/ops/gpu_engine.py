@@ -278,6 +278,54 @@ def _nodes_packed_v4(forest, d_sentinel: int, bf16: bool):
     return packed, ncount, max(max_depth, 1)
 
 
+def _nodes_packed_wide(forest, bf16: bool):
+    """Wide int4 node records {feat, right, key_or_value, 0} for
+    score_forest_wide — no 12-bit feature / 15-bit node-id limits
+    (the reference has no such size caps,
+    IsolationTree.scala). Same key/leaf-value math as _nodes_packed_v4,
+    so the walk stays bitwise vs cpu_engine.path_lengths. Leaves and pad
+    nodes self-loop (right = own id)."""
+    T, mn = forest.feature.shape
+    feat = forest.feature
+    right = forest.right
+    val = forest.value
+    internal = feat >= 0
+    leaf = feat == Forest.LEAF
+    depth = _node_depths(feat, right)
+    ids = np.broadcast_to(np.arange(mn, dtype=np.int32)[None, :], (T, mn))
+    w0 = np.where(internal, feat.astype(np.int32), np.int32(-1))
+    w1 = np.where(internal, right.astype(np.int32), ids).astype(np.int32)
+    w2 = np.zeros((T, mn), dtype=np.uint32)
+    if internal.any():
+        s = val[internal].astype(np.float32)
+        keys = _bf16_threshold_keys(s) if bf16 else _key32(s)
+        w2[internal] = keys.astype(np.uint32)
+    leafval = depth.astype(np.float32) + val.astype(np.float32)
+    w2[leaf] = leafval[leaf].astype(np.float32).view(np.uint32)
+    packed = np.zeros((T, mn, 4), dtype=np.int32)
+    packed[..., 0] = w0
+    packed[..., 1] = w1
+    packed[..., 2] = w2.view(np.int32)
+    live = internal | leaf
+    max_depth = int(depth[live].max()) if live.any() else 0
+    return packed, max(max_depth, 1)
+
+
+def _eif_packed_wide(forest):
+    """Wide int4 node records {is_leaf(-1)/0, right, value_bits, 0} for
+    score_extended_wide (strict oracle j-order dot from global
+    hyperplanes)."""
+    T, mn = forest.feature.shape
+    feat = forest.feature
+    internal = feat >= 0
+    ids = np.broadcast_to(np.arange(mn, dtype=np.int32)[None, :], (T, mn))
+    packed = np.zeros((T, mn, 4), dtype=np.int32)
+    packed[..., 0] = np.where(internal, np.int32(0), np.int32(-1))
+    packed[..., 1] = np.where(internal, forest.right.astype(np.int32), ids)
+    packed[..., 2] = forest.value.astype(np.float32).view(np.int32)
+    return packed
+
+
 def _densify_weights(hidx: np.ndarray, hw: np.ndarray, counts: np.ndarray,
                      D: int) -> np.ndarray:
     """Scatter sparse hyperplanes [T, mn, nnz] into dense rows [T, mn, D].
@@ -538,6 +586,23 @@ def _device_forest(model, device, v4_key=None):
                     hb[..., 0::2] | (hb[..., 1::2] << 16)
                 ).contiguous()
             extra["height"] = max_depth
+        elif isinstance(v4_key, tuple) and v4_key[0] == "wide":
+            packed, max_depth = _nodes_packed_wide(forest, v4_key[1])
+            aos = torch.from_numpy(packed).to(device)
+            ncount = torch.from_numpy(
+                np.ascontiguousarray(forest.node_count, dtype=np.int32)
+            ).to(device)
+            extra["height"] = max_depth
+        elif v4_key == "eif_wide":
+            packed = _eif_packed_wide(forest)
+            aos = torch.from_numpy(packed).to(device)
+            ncount = torch.from_numpy(
+                np.ascontiguousarray(forest.node_count, dtype=np.int32)
+            ).to(device)
+            extra["hidx"] = torch.from_numpy(
+                np.ascontiguousarray(forest.hyper_idx)).to(device)
+            extra["hw"] = torch.from_numpy(
+                np.ascontiguousarray(forest.hyper_w)).to(device)
         elif v4_key is not None:
             d, bf16 = v4_key
             if raw is not None:
@@ -568,13 +633,30 @@ def _device_forest(model, device, v4_key=None):
     return cache[key]
 
 
+def _packed_fits(forest, d: int) -> bool:
+    """True when the packed v4/EIF formats can represent this forest
+    (12-bit feature ids, 15-bit node ids)."""
+    mn = forest.feature.shape[1]
+    if mn > 32767:
+        return False
+    return int(forest.feature.max(initial=0)) <= 4094 and d <= 4094
+
+
 def score_forest(model, X: torch.Tensor, finalize: bool = True) -> torch.Tensor:
     ext = load_extension()
     forest = model.forest
     d = int(X.shape[1])
     bf16 = X.dtype == torch.bfloat16
-    aos, ncount, extra = _device_forest(model, X.device, v4_key=(d, bf16))
     c = float(avg_path_length(forest.num_samples))
+    if not _packed_fits(forest, d):
+        # loaded/CPU-built forests past the packed-format caps: wide int4
+        # records, no field-width limits (VERDICT r01 #5)
+        aos, ncount, extra = _device_forest(
+            model, X.device, v4_key=("wide", bf16))
+        return ext.score_forest_wide(
+            X.contiguous(), aos, extra["height"], c, finalize,
+        )
+    aos, ncount, extra = _device_forest(model, X.device, v4_key=(d, bf16))
     return ext.score_forest(
         X.contiguous(), aos, ncount, forest.num_trees, extra["height"], c,
         finalize,
@@ -601,6 +683,14 @@ def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torc
     c = float(avg_path_length(forest.num_samples))
     d = int(X.shape[1])
     nnz = forest.nnz
+    mn = forest.feature.shape[1]
+    if mn > 32767 or int(forest.feature.max(initial=0)) > 4094:
+        # past the packed 15-bit node-id / 12-bit count caps: wide records
+        aos, ncount, extra = _device_forest(model, X.device,
+                                            v4_key="eif_wide")
+        return ext.score_extended_wide(
+            X.contiguous(), aos, extra["hidx"], extra["hw"], c, finalize,
+        )
     # routing (measured crossovers, profiles/r01_bench_and_kernels.md):
     # nnz <= 5 with uniform hyperplane widths -> fixed-trip sparse v2;
     # wider hyperplanes with d <= 32 -> densified dense v2 walk;

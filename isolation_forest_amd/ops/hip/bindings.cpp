@@ -86,6 +86,19 @@ void launch_score_extended_dense_v3(int D, const void* X, const void* nodes,
                                     size_t lds, int blocks,
                                     hipStream_t stream);
 
+void launch_score_forest_wide(bool bf16, const void* X, const void* nodes,
+                              float* out, int64_t N, int32_t d, int32_t T,
+                              int64_t max_nodes, int32_t height_limit,
+                              float fT, float c_norm, int finalize,
+                              int blocks, hipStream_t stream);
+
+void launch_score_extended_wide(bool bf16, const void* X, const void* nodes,
+                                const int32_t* hidx, const float* hw,
+                                float* out, int64_t N, int32_t d, int32_t T,
+                                int64_t max_nodes, int32_t nnz, float fT,
+                                float c_norm, int finalize, int blocks,
+                                hipStream_t stream);
+
 }  // namespace ifa
 
 namespace {
@@ -485,6 +498,59 @@ torch::Tensor score_extended_sparse_v2(torch::Tensor X,
   return out;
 }
 
+torch::Tensor score_forest_wide(torch::Tensor X, torch::Tensor nodes_wide,
+                                int64_t height_limit, double c_norm,
+                                bool finalize) {
+  CHECK_CUDA(X);
+  CHECK_CONTIG(X);
+  CHECK_CUDA(nodes_wide);
+  CHECK_CONTIG(nodes_wide);
+  check_x(X);
+  TORCH_CHECK(nodes_wide.dim() == 3 && nodes_wide.size(2) == 4 &&
+                  nodes_wide.scalar_type() == torch::kInt32,
+              "nodes must be wide int32 [T, max_nodes, 4]");
+  int64_t N = X.size(0), d = X.size(1);
+  int64_t T = nodes_wide.size(0), max_nodes = nodes_wide.size(1);
+  auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
+  if (N == 0) return out;
+  int blocks = (int)std::min<int64_t>((N + 255) / 256, 8192);
+  ifa::launch_score_forest_wide(
+      is_bf16(X), X.data_ptr(), nodes_wide.data_ptr<int32_t>(),
+      out.data_ptr<float>(), N, (int32_t)d, (int32_t)T, max_nodes,
+      (int32_t)height_limit, (float)T, (float)c_norm, finalize ? 1 : 0,
+      blocks, current_stream());
+  return out;
+}
+
+torch::Tensor score_extended_wide(torch::Tensor X, torch::Tensor nodes_wide,
+                                  torch::Tensor hidx, torch::Tensor hw,
+                                  double c_norm, bool finalize) {
+  CHECK_CUDA(X);
+  CHECK_CONTIG(X);
+  CHECK_CUDA(nodes_wide);
+  CHECK_CONTIG(nodes_wide);
+  CHECK_CUDA(hidx);
+  CHECK_CONTIG(hidx);
+  CHECK_CUDA(hw);
+  CHECK_CONTIG(hw);
+  check_x(X);
+  TORCH_CHECK(nodes_wide.dim() == 3 && nodes_wide.size(2) == 4 &&
+                  nodes_wide.scalar_type() == torch::kInt32,
+              "nodes must be wide int32 [T, max_nodes, 4]");
+  int64_t N = X.size(0), d = X.size(1);
+  int64_t T = nodes_wide.size(0), max_nodes = nodes_wide.size(1);
+  int64_t nnz = hidx.size(2);
+  auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
+  if (N == 0) return out;
+  int blocks = (int)std::min<int64_t>((N + 255) / 256, 8192);
+  ifa::launch_score_extended_wide(
+      is_bf16(X), X.data_ptr(), nodes_wide.data_ptr<int32_t>(),
+      hidx.data_ptr<int32_t>(), hw.data_ptr<float>(), out.data_ptr<float>(),
+      N, (int32_t)d, (int32_t)T, max_nodes, (int32_t)nnz, (float)T,
+      (float)c_norm, finalize ? 1 : 0, blocks, current_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bag_gather", &bag_gather, "gather per-tree bags (K9/K10)");
   m.def("build_forest", &build_forest, "build standard iTrees (K1/K2/K11)");
@@ -499,5 +565,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "dense EIF scoring, rows-in-registers (K7 fast path)");
   m.def("score_extended_dense_v3", &score_extended_dense_v3,
         "dense EIF scoring, packed-bf16 weights + v_dot2c (K7 bf16 path)");
+  m.def("score_forest_wide", &score_forest_wide,
+        "wide-format standard scoring (no packed-field limits)");
+  m.def("score_extended_wide", &score_extended_wide,
+        "wide-format EIF scoring (no packed-field limits)");
   m.attr("WAVE") = 64;
 }

@@ -1300,6 +1300,99 @@ __global__ void __launch_bounds__(256) score_extended_forest_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// WIDE-format fallback scoring (no packed-field width limits).
+//
+// The packed v4/EIF node formats cap feature ids at 12 bits (4094) and
+// node ids at 15 bits (32767 nodes/tree = maxSamples 16384). Forests
+// loaded from foreign files (or CPU-built past the GPU build cap) can
+// exceed both; these kernels store each node as a full int4
+// {feat, right, key/value, 0} read from GLOBAL memory (L2-served: every
+// workgroup walks trees in the same order) with rows read from global —
+// a correctness path, not a throughput path (the reference has no such
+// size limits: IsolationTree.scala). Same integer-key compare math as v4
+// => bitwise vs cpu_engine.path_lengths.
+// ---------------------------------------------------------------------------
+
+template <typename KT>
+__global__ void __launch_bounds__(256) score_forest_wide(
+    const KT* __restrict__ X,          // raw bf16/f32 bits [N][d]
+    const int4* __restrict__ nodes,    // [T][max_nodes] {feat, right, w, 0}
+    float* __restrict__ out, int64_t N, int32_t d, int32_t T,
+    int64_t max_nodes, int32_t height_limit, float fT, float c_norm,
+    int32_t finalize) {
+  for (int64_t my_row = (int64_t)blockIdx.x * 256 + threadIdx.x; my_row < N;
+       my_row += (int64_t)gridDim.x * 256) {
+    float psum = 0.f;
+    for (int t = 0; t < T; ++t) {
+      const int4* base = nodes + (int64_t)t * max_nodes;
+      int cur = 0;
+      for (int it = 0; it < height_limit; ++it) {
+        const int4 nd = base[cur];
+        if (nd.y != cur) {  // leaves self-loop (right == own id)
+          const uint32_t x =
+              widen_key(key_of_bits<KT>(X[my_row * d + nd.x]));
+          cur = (x < (uint32_t)nd.z) ? cur + 1 : nd.y;
+        }
+      }
+      psum = __fadd_rn(psum, __int_as_float(base[cur].z));
+    }
+    if (finalize) {
+      const float mean32 = (float)((double)psum / (double)fT);
+      const double ratio = (double)mean32 / (double)c_norm;
+      out[my_row] = (float)exp2(-ratio);
+    } else {
+      out[my_row] = psum;
+    }
+  }
+}
+
+// EIF wide fallback: full-width node records, strict oracle j-order dot
+// from global hyperplanes (bitwise vs cpu_engine.path_lengths_extended).
+template <typename XT>
+__global__ void __launch_bounds__(256) score_extended_wide(
+    const XT* __restrict__ X, const int4* __restrict__ nodes,
+    const int32_t* __restrict__ hidx_g,  // [T][max_nodes][nnz]
+    const float* __restrict__ hw_g,      // [T][max_nodes][nnz]
+    float* __restrict__ out, int64_t N, int32_t d, int32_t T,
+    int64_t max_nodes, int32_t nnz, float fT, float c_norm,
+    int32_t finalize) {
+  for (int64_t my_row = (int64_t)blockIdx.x * 256 + threadIdx.x; my_row < N;
+       my_row += (int64_t)gridDim.x * 256) {
+    float path_sum = 0.f;
+    for (int t = 0; t < T; ++t) {
+      const int4* base = nodes + (int64_t)t * max_nodes;
+      const int64_t hbase = (int64_t)t * max_nodes * nnz;
+      int cur = 0, dep = 0;
+      float leaf = 0.f;
+      while (true) {
+        const int4 nd = base[cur];
+        if (nd.x < 0) {
+          leaf = __int_as_float(nd.z);
+          break;
+        }
+        float dot = 0.f;
+        const int32_t* ci = hidx_g + hbase + (int64_t)cur * nnz;
+        const float* cw = hw_g + hbase + (int64_t)cur * nnz;
+        for (int j = 0; j < nnz; ++j) {
+          const float xv = load_feat<XT>(X, my_row * d + ci[j]);
+          dot = __fadd_rn(dot, __fmul_rn(cw[j], xv));
+        }
+        cur = (dot < __int_as_float(nd.z)) ? cur + 1 : nd.y;
+        ++dep;
+      }
+      path_sum = __fadd_rn(path_sum, __fadd_rn((float)dep, leaf));
+    }
+    if (finalize) {
+      const float mean32 = (float)((double)path_sum / (double)fT);
+      const double ratio = (double)mean32 / (double)c_norm;
+      out[my_row] = (float)exp2(-ratio);
+    } else {
+      out[my_row] = path_sum;
+    }
+  }
+}
+
 }  // namespace ifa
 
 // ---------------------------------------------------------------------------
@@ -1460,6 +1553,39 @@ void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
     else LSD2(uint32_t, 32, 1);
   }
 #undef LSD2
+}
+
+void launch_score_forest_wide(bool bf16, const void* X, const void* nodes,
+                              float* out, int64_t N, int32_t d, int32_t T,
+                              int64_t max_nodes, int32_t height_limit,
+                              float fT, float c_norm, int finalize,
+                              int blocks, hipStream_t stream) {
+  if (bf16)
+    hipLaunchKernelGGL((score_forest_wide<uint16_t>), dim3(blocks), dim3(256),
+                       0, stream, (const uint16_t*)X, (const int4*)nodes, out,
+                       N, d, T, max_nodes, height_limit, fT, c_norm, finalize);
+  else
+    hipLaunchKernelGGL((score_forest_wide<uint32_t>), dim3(blocks), dim3(256),
+                       0, stream, (const uint32_t*)X, (const int4*)nodes, out,
+                       N, d, T, max_nodes, height_limit, fT, c_norm, finalize);
+}
+
+void launch_score_extended_wide(bool bf16, const void* X, const void* nodes,
+                                const int32_t* hidx, const float* hw,
+                                float* out, int64_t N, int32_t d, int32_t T,
+                                int64_t max_nodes, int32_t nnz, float fT,
+                                float c_norm, int finalize, int blocks,
+                                hipStream_t stream) {
+  if (bf16)
+    hipLaunchKernelGGL((score_extended_wide<uint16_t>), dim3(blocks),
+                       dim3(256), 0, stream, (const uint16_t*)X,
+                       (const int4*)nodes, hidx, hw, out, N, d, T, max_nodes,
+                       nnz, fT, c_norm, finalize);
+  else
+    hipLaunchKernelGGL((score_extended_wide<float>), dim3(blocks), dim3(256),
+                       0, stream, (const float*)X, (const int4*)nodes, hidx,
+                       hw, out, N, d, T, max_nodes, nnz, fT, c_norm,
+                       finalize);
 }
 
 void launch_score_extended_dense_v3(int D, const void* X, const void* nodes,

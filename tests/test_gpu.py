@@ -598,6 +598,69 @@ class TestDeepForests:
             s_gpu.view(np.int32), s_cpu.view(np.int32))
 
 
+class TestWideFormatFallbacks:
+    """Beyond the packed-format caps (15-bit node ids => maxSamples 16384,
+    12-bit feature ids => d 4094) the wide int4 kernels take over —
+    VERDICT r01 #5: no config that fits can fail to score. Boundary on
+    both sides, bitwise contracts."""
+
+    def test_build_cap_boundary_16384_gpu_build(self, dev):
+        X = torch.from_numpy(make_data(20000, 6, seed=70)).to(dev)
+        model = IsolationForest(
+            numEstimators=4, maxSamples=16384.0, randomSeed=19).fit(X)
+        assert model.forest.num_samples == 16384
+        s_gpu = model.score(X).cpu().numpy()
+        s_cpu = model.score(X.float().cpu()).numpy()
+        np.testing.assert_array_equal(
+            s_gpu.view(np.int32), s_cpu.view(np.int32))
+
+    def test_build_cap_boundary_16385_cpu_fallback(self, dev):
+        """One past the GPU build cap: fit must still succeed on a CUDA
+        tensor (CPU tree build + GPU wide scoring) and match the CPU
+        model bitwise."""
+        X = torch.from_numpy(make_data(20000, 6, seed=71)).to(dev)
+        model = IsolationForest(
+            numEstimators=2, maxSamples=16385.0, randomSeed=20).fit(X)
+        assert model.forest.num_samples == 16385
+        assert model.forest.feature.shape[1] > 32767  # wide territory
+        s_gpu = model.score(X).cpu().numpy()
+        s_cpu = model.score(X.float().cpu()).numpy()
+        np.testing.assert_array_equal(
+            s_gpu.view(np.int32), s_cpu.view(np.int32))
+
+    def test_wide_d_4095_feature_ids(self, dev):
+        """d = 4095 puts feature id 4094+ past the 12-bit packed field:
+        routes to score_forest_wide, bitwise vs the oracle."""
+        rs = np.random.RandomState(72)
+        X = rs.normal(size=(2000, 4095)).astype(np.float32)
+        bag = cpu_engine.sample_bags(2000, 3, 256, seed=21, bootstrap=False)
+        fs = cpu_engine.feature_subsets(4095, 4095, 3, seed=21)
+        forest = cpu_engine.build_forest(X, bag, fs, 21, 256, 4095, 4095)
+        cpu_ps = cpu_engine.path_lengths(forest, X)
+        model = IsolationForest(numEstimators=3).fit(X[:500])
+        model.forest = forest
+        model._gpu_forest_cache = {}
+        from isolation_forest_amd.ops import gpu_engine
+
+        assert not gpu_engine._packed_fits(forest, 4095)
+        gpu_ps = gpu_engine.score_forest(
+            model, torch.from_numpy(X).to(dev), finalize=False)
+        np.testing.assert_array_equal(
+            gpu_ps.cpu().numpy().view(np.int32), cpu_ps.view(np.int32))
+
+    def test_extended_wide_nodes(self, dev):
+        """EIF forest past 32767 nodes/tree: score_extended_wide keeps the
+        strict j-order dot (bitwise)."""
+        X = torch.from_numpy(make_data(25000, 5, seed=73)).to(dev)
+        model = ExtendedIsolationForest(
+            numEstimators=2, maxSamples=17000.0, randomSeed=22).fit(X)
+        assert model.forest.feature.shape[1] > 32767
+        s_gpu = model.score(X).cpu().numpy()
+        s_cpu = model.score(X.float().cpu()).numpy()
+        np.testing.assert_array_equal(
+            s_gpu.view(np.int32), s_cpu.view(np.int32))
+
+
 class TestParityFuzzSmoke:
     def test_ten_random_configs(self, dev):
         import subprocess

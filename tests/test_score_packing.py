@@ -198,3 +198,68 @@ class TestThresholdKeyClosedForm:
         s = s[~np.isnan(s)]
         np.testing.assert_array_equal(
             _bf16_threshold_keys(s), _bf16_threshold_keys_table(s))
+
+
+class TestWidePacking:
+    """Wide int4 node records (score_forest_wide / score_extended_wide):
+    same key/leaf-value math as the packed v4 format, fields unpacked."""
+
+    def _forest(self):
+        rs = np.random.RandomState(9)
+        X = rs.normal(size=(3000, 6)).astype(np.float32)
+        bag = cpu_engine.sample_bags(3000, 8, 256, seed=4, bootstrap=False)
+        fs = cpu_engine.feature_subsets(6, 6, 8, seed=4)
+        return cpu_engine.build_forest(X, bag, fs, 4, 256, 6, 6)
+
+    def test_fields_match_v4_semantics(self):
+        from isolation_forest_amd.ops.gpu_engine import (
+            _nodes_packed_v4, _nodes_packed_wide)
+
+        forest = self._forest()
+        for bf16 in (False, True):
+            v4, _, h4 = _nodes_packed_v4(forest, 6, bf16)
+            wide, hw_ = _nodes_packed_wide(forest, bf16)
+            assert hw_ == h4
+            T, mn = forest.feature.shape
+            internal = forest.feature >= 0
+            leaf = forest.feature == -1
+            # internal: feat/right fields equal the v4 bit fields
+            np.testing.assert_array_equal(
+                wide[..., 0][internal], v4[:T][..., 0][internal] & 0xFFF)
+            np.testing.assert_array_equal(
+                wide[..., 1][internal],
+                (v4[:T][..., 0][internal] >> 12) & 0x7FFF)
+            # keys/leaf values identical
+            np.testing.assert_array_equal(
+                wide[..., 2][internal | leaf],
+                v4[:T][..., 1][internal | leaf])
+            # leaves and pad self-loop
+            ids = np.broadcast_to(np.arange(mn, dtype=np.int32), (T, mn))
+            np.testing.assert_array_equal(wide[..., 1][~internal],
+                                          ids[~internal])
+
+    def test_packed_fits_predicate(self):
+        from isolation_forest_amd.ops.gpu_engine import _packed_fits
+
+        forest = self._forest()
+        assert _packed_fits(forest, 6)
+        assert _packed_fits(forest, 4094)
+        assert not _packed_fits(forest, 4095)
+
+    def test_eif_wide_fields(self):
+        from isolation_forest_amd.ops.gpu_engine import _eif_packed_wide
+
+        rs = np.random.RandomState(10)
+        X = rs.normal(size=(2000, 5)).astype(np.float32)
+        bag = cpu_engine.sample_bags(2000, 4, 128, seed=5, bootstrap=False)
+        fs = cpu_engine.feature_subsets(5, 5, 4, seed=5)
+        forest = cpu_engine.build_extended_forest(X, bag, fs, 5, 128, 5, 5, 4)
+        wide = _eif_packed_wide(forest)
+        internal = forest.feature >= 0
+        assert (wide[..., 0][internal] == 0).all()
+        assert (wide[..., 0][~internal] == -1).all()
+        np.testing.assert_array_equal(
+            wide[..., 1][internal], forest.right[internal])
+        np.testing.assert_array_equal(
+            wide[..., 2].view(np.float32)[internal],
+            forest.value.astype(np.float32)[internal])
