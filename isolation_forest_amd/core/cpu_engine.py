@@ -431,9 +431,18 @@ def path_lengths(forest: Forest, X: np.ndarray) -> np.ndarray:
     """
     N = X.shape[0]
     total = np.zeros(N, dtype=np.float32)
+    # splits compare in FLOAT64 against the exact persisted values, matching
+    # the reference's Float-promoted-to-Double compare
+    # (IsolationTree.scala:213-229): x < splitValue with splitValue a Double.
+    # For our own forests value64 is exactly the f32 split, so this is
+    # bitwise-identical to an f32 compare; for loaded foreign models it
+    # removes the knife-edge flips f32 rounding produced.
+    v64_all = (forest.value64 if getattr(forest, "value64", None) is not None
+               else forest.value.astype(np.float64))
     for t in range(forest.num_trees):
         feat = forest.feature[t]
         val = forest.value[t]
+        v64 = v64_all[t]
         right = forest.right[t]
         cur = np.zeros(N, dtype=np.int64)
         depth = np.zeros(N, dtype=np.int32)
@@ -441,7 +450,7 @@ def path_lengths(forest: Forest, X: np.ndarray) -> np.ndarray:
         while active.any():
             f = feat[cur[active]]
             x = X[np.nonzero(active)[0], f]
-            go_left = x < val[cur[active]]
+            go_left = x < v64[cur[active]]
             nxt = np.where(go_left, cur[active] + 1, right[cur[active]])
             cur[active] = nxt
             depth[active] += 1

@@ -24,6 +24,8 @@ import json
 import math
 from typing import Dict
 
+import numpy as np
+
 from ..persist import avro_io
 from . import model_proto as mp
 
@@ -212,7 +214,18 @@ class IsolationForestConverter:
         attrs["nodes_nodeids"].append(node_id)
         attrs["nodes_featureids"].append(int(nd["splitAttribute"]))
         attrs["nodes_modes"].append("LEAF" if is_leaf else "BRANCH_LT")
-        attrs["nodes_values"].append(float(nd["splitValue"]))
+        # nodes_values is an f32 FLOATS attribute but the persisted
+        # splitValue is a double; BRANCH_LT tests x < value, and for f32 x
+        # and f64 s: x < s <=> x < ceil32(s) (smallest f32 >= s). Emitting
+        # the ceil-rounded value makes ONNX inference reproduce the
+        # engine's (and Spark's) f64 compare EXACTLY — round-to-nearest
+        # (what the reference converter implicitly does) flips knife-edge
+        # rows (PARITY.md "Foreign-model f64 split semantics").
+        s64 = float(nd["splitValue"])
+        s32 = np.float32(s64)
+        if not is_leaf and float(s32) < s64:
+            s32 = np.nextafter(s32, np.float32(np.inf))
+        attrs["nodes_values"].append(float(s32) if not is_leaf else s64)
         attrs["nodes_truenodeids"].append(left)
         attrs["nodes_falsenodeids"].append(right)
         attrs["nodes_missing_value_tracks_true"].append(0)

@@ -231,6 +231,23 @@ def _node_depths(feature: np.ndarray, right: np.ndarray) -> np.ndarray:
     return depth
 
 
+def _split_thresholds32(forest) -> np.ndarray:
+    """f32 thresholds implementing the EXACT f64 split compare.
+
+    The reference compares (double)(float x) < splitValue with splitValue
+    a double (IsolationTree.scala:213-229). For f32 x and f64 s:
+    x < s  <=>  x < ceil32(s) (the smallest f32 >= s), because no f32 lies
+    strictly between two f32 neighbours. Our own builds store f32-exact
+    splits (value64 == f32 value), so ceil32 is the identity there; loaded
+    foreign models get the exact Spark semantics (VERDICT r01 #6)."""
+    v64 = (forest.value64 if getattr(forest, "value64", None) is not None
+           else forest.value.astype(np.float64))
+    v32 = v64.astype(np.float32)
+    lift = v32.astype(np.float64) < v64
+    return np.where(lift, np.nextafter(v32, np.float32(np.inf)),
+                    v32).astype(np.float32)
+
+
 def _nodes_packed_v4(forest, d_sentinel: int, bf16: bool):
     """v4 packed nodes for score_forest_v4 (see forest_kernels.hip):
     w0 = feat | right<<12; internal w1 = integer key threshold; leaf:
@@ -257,7 +274,7 @@ def _nodes_packed_v4(forest, d_sentinel: int, bf16: bool):
     ).astype(np.int32)
     w1 = np.zeros((T, mn), dtype=np.uint32)
     if internal.any():
-        s = val[internal].astype(np.float32)
+        s = _split_thresholds32(forest)[internal]
         w1[internal] = _bf16_threshold_keys(s) if bf16 else _key32(s)
     leafval = (depth.astype(np.float32) + val.astype(np.float32))
     w1[leaf] = leafval[leaf].astype(np.float32).view(np.uint32)
@@ -297,7 +314,7 @@ def _nodes_packed_wide(forest, bf16: bool):
     w1 = np.where(internal, right.astype(np.int32), ids).astype(np.int32)
     w2 = np.zeros((T, mn), dtype=np.uint32)
     if internal.any():
-        s = val[internal].astype(np.float32)
+        s = _split_thresholds32(forest)[internal]
         keys = _bf16_threshold_keys(s) if bf16 else _key32(s)
         w2[internal] = keys.astype(np.uint32)
     leafval = depth.astype(np.float32) + val.astype(np.float32)

@@ -598,6 +598,38 @@ class TestDeepForests:
             s_gpu.view(np.int32), s_cpu.view(np.int32))
 
 
+class TestForeignF64SplitsGPU:
+    def test_knife_edge_foreign_model_bitwise(self, dev):
+        """Foreign f64 splits strictly between adjacent f32s, rows planted
+        ON the boundary: GPU ceil32 threshold keys must implement the
+        exact f64 compare, bitwise vs the (f64-compare) CPU engine."""
+        from isolation_forest_amd.ops import gpu_engine
+
+        rs = np.random.RandomState(31)
+        X = rs.normal(size=(2500, 4)).astype(np.float32)
+        bag = cpu_engine.sample_bags(2500, 8, 128, seed=31, bootstrap=False)
+        fs = cpu_engine.feature_subsets(4, 4, 8, seed=31)
+        forest = cpu_engine.build_forest(X, bag, fs, 31, 128, 4, 4)
+        internal = forest.feature >= 0
+        v32 = forest.value.astype(np.float64)
+        nxt = np.nextafter(forest.value,
+                           np.float32(np.inf)).astype(np.float64)
+        forest.value64 = np.where(internal, (v32 + nxt) / 2.0,
+                                  forest.value64)
+        ik = np.argwhere(internal)
+        pick = ik[rs.randint(0, len(ik), size=600)]
+        for r, (t, n) in enumerate(pick):
+            X[r, forest.feature[t, n]] = forest.value[t, n]
+        cpu_ps = cpu_engine.path_lengths(forest, X)
+        model = IsolationForest(numEstimators=8).fit(X[:500])
+        model.forest = forest
+        model._gpu_forest_cache = {}
+        gpu_ps = gpu_engine.score_forest(
+            model, torch.from_numpy(X).to(dev), finalize=False)
+        np.testing.assert_array_equal(
+            gpu_ps.cpu().numpy().view(np.int32), cpu_ps.view(np.int32))
+
+
 class TestWideFormatFallbacks:
     """Beyond the packed-format caps (15-bit node ids => maxSamples 16384,
     12-bit feature ids => d 4094) the wide int4 kernels take over —

@@ -431,7 +431,11 @@ static Packed pack_v4(const Model& m, int d_sentinel, bool bf16) {
         p.max_depth = std::max(p.max_depth, depth.at(nd.id));
       } else {
         w0 = nd.split_attr | (nd.right << 12);
+        // exact f64 compare semantics: x < s(f64) <=> x < ceil32(s)
+        // (IsolationTree.scala:213-229; see gpu_engine._split_thresholds32)
         float s32 = (float)nd.split_value;
+        if ((double)s32 < nd.split_value)
+          s32 = std::nextafterf(s32, INFINITY);
         uint32_t key = bf16 ? bf16_threshold_key(s32) : f32_key(s32);
         memcpy(&w1, &key, 4);
         p.max_depth = std::max(p.max_depth, depth.at(nd.id) + 1);
