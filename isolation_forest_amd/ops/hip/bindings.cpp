@@ -77,8 +77,9 @@ void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
                                     float c_norm, int finalize, size_t lds,
                                     int blocks, hipStream_t stream);
 
-void launch_score_extended_dense_v3(int D, const void* X, const void* nodes,
-                                    const float* values, const uint32_t* hwp,
+void launch_score_extended_dense_v3(int D, bool rpt2, const void* X,
+                                    const void* nodes, const float* values,
+                                    const uint32_t* hwp,
                                     const int32_t* ncount, float* out,
                                     int64_t N, int32_t d, int32_t T,
                                     int32_t max_nodes, int32_t height_limit,
@@ -440,9 +441,13 @@ torch::Tensor score_extended_dense_v3(torch::Tensor X,
 
   size_t lds = (size_t)max_nodes * 12 + 16 + (size_t)max_nodes * (D / 8 + 1) * 16;
   TORCH_CHECK(lds <= kMaxLds, "tree too large for LDS staging");
-  int blocks = (int)std::min<int64_t>((N + 511) / 512, 8192);
+  const char* rpt_env = getenv("IFA_EIF_V3_RPT2");
+  const bool rpt2 = rpt_env && rpt_env[0] == '1';
+  const int rows_per_iter = (D == 32 && !rpt2) ? 512 : 1024;
+  int blocks = (int)std::min<int64_t>(
+      (N + rows_per_iter - 1) / rows_per_iter, 8192);
   ifa::launch_score_extended_dense_v3(
-      D, X.data_ptr(), nodes_packed.data_ptr<int32_t>(),
+      D, rpt2, X.data_ptr(), nodes_packed.data_ptr<int32_t>(),
       values.data_ptr<float>(), (const uint32_t*)hwp.data_ptr<int32_t>(),
       ncount.data_ptr<int32_t>(), out.data_ptr<float>(), N, (int32_t)d,
       (int32_t)T, (int32_t)max_nodes, (int32_t)height_limit, (float)T,

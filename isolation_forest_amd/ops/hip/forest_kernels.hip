@@ -971,7 +971,9 @@ __device__ __forceinline__ float dot2_bf16(uint32_t w, uint32_t x, float acc) {
 }
 
 template <int D, int RPT>
-__global__ void __launch_bounds__(EIFD_THREADS, 6) score_extended_dense_v3(
+__global__ void __launch_bounds__(EIFD_THREADS,
+                                  (D == 32 && RPT == 2) ? 4 : 6)
+score_extended_dense_v3(
     const uint16_t* __restrict__ X,     // raw bf16 bits [N][d]
     const int2* __restrict__ nodes,     // [T][max_nodes] {right<<12, offset/-inf}
     const float* __restrict__ values,   // [T][max_nodes] leaf value+depth
@@ -1588,8 +1590,9 @@ void launch_score_extended_wide(bool bf16, const void* X, const void* nodes,
                        finalize);
 }
 
-void launch_score_extended_dense_v3(int D, const void* X, const void* nodes,
-                                    const float* values, const uint32_t* hwp,
+void launch_score_extended_dense_v3(int D, bool rpt2, const void* X,
+                                    const void* nodes, const float* values,
+                                    const uint32_t* hwp,
                                     const int32_t* ncount, float* out,
                                     int64_t N, int32_t d, int32_t T,
                                     int32_t max_nodes, int32_t height_limit,
@@ -1606,6 +1609,7 @@ void launch_score_extended_dense_v3(int D, const void* X, const void* nodes,
   } while (0)
   if (D == 8) LSD3(8, 2);
   else if (D == 16) LSD3(16, 2);
+  else if (rpt2) LSD3(32, 2);  // A/B: halves per-row staging amortization
   else LSD3(32, 1);
 #undef LSD3
 }
