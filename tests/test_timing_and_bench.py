@@ -115,3 +115,12 @@ class TestTimingSync:
                 pass
         assert times["p"] >= 0.0
         assert times.total == sum(times.values())
+
+
+class TestOfflineLint:
+    def test_repo_is_lint_clean(self):
+        out = subprocess.run(
+            [sys.executable, "tools/lint_offline.py"],
+            cwd=REPO, capture_output=True, text=True, timeout=120,
+        )
+        assert out.returncode == 0, out.stdout[-3000:]
