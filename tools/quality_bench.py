@@ -9,8 +9,8 @@ EIF fully-extended, on CPU or GPU.
 
     python tools/quality_bench.py [--trials 10] [--device cpu|cuda:0]
 
-Requires the reference checkout's CSVs (present in the dev container at
-/root/reference); prints a markdown table for profiles/.
+Datasets load from the committed in-repo fixtures (tests/fixtures/*.npz)
+so this runs on GPU boxes too; prints a markdown table for profiles/.
 """
 import argparse
 import os
@@ -22,7 +22,8 @@ import torch
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from isolation_forest_amd import ExtendedIsolationForest, IsolationForest
 
-RES = "/root/reference/isolation-forest/src/test/resources"
+FIXTURES = os.path.join(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))), "tests", "fixtures")
 
 PUBLISHED = {  # BASELINE.md (README.md:441-443, 453-455)
     ("mammography", "StandardIF"): (0.8649, 0.0015),
@@ -44,9 +45,8 @@ def auroc(y, s):
 
 
 def load(name):
-    path = os.path.join(RES, f"{name}.csv")
-    data = np.loadtxt(path, delimiter=",", comments="#")
-    return data[:, :-1].astype(np.float32), data[:, -1]
+    with np.load(os.path.join(FIXTURES, f"{name}.npz")) as z:
+        return z["X"].astype(np.float32), z["y"].astype(np.float64)
 
 
 def run(name, X, y, trials, device):
@@ -83,8 +83,8 @@ def main():
     ap.add_argument("--datasets", nargs="+",
                     default=["mammography", "shuttle"])
     args = ap.parse_args()
-    if not os.path.isdir(RES):
-        print("reference resources not found; skipping")
+    if not os.path.isdir(FIXTURES):
+        print("fixtures not found; skipping")
         return
     print("| dataset | model | this engine (mean ± SEM) | reference published |")
     print("|---|---|---|---|")
