@@ -705,3 +705,26 @@ class TestParityFuzzSmoke:
              "--iters", "10", "--seed", "42"],
             capture_output=True, text=True, timeout=600, cwd=repo)
         assert out.returncode == 0, out.stdout + out.stderr
+
+
+class TestServingGPU:
+    def test_http_scoring_runs_hip_kernels(self, dev, tmp_path, mammography):
+        """The serving layer on a CUDA device must score through the same
+        HIP path as batch transform (parity at float tolerance)."""
+        from fastapi.testclient import TestClient
+
+        from isolation_forest_amd.serving import create_app
+
+        X, y = mammography
+        model = IsolationForest(
+            numEstimators=50, contamination=0.02, randomSeed=9
+        ).fit(torch.from_numpy(X).to(dev))
+        path = str(tmp_path / "m")
+        model.save(path)
+        client = TestClient(create_app(path, device=str(dev)))
+        assert client.get("/v1/model").json()["device"] == str(dev)
+        r = client.post("/v1/score", json={"instances": X[:256].tolist()})
+        assert r.status_code == 200
+        served = np.asarray(r.json()["scores"], dtype=np.float32)
+        engine = model.score(torch.from_numpy(X[:256]).to(dev)).cpu().numpy()
+        np.testing.assert_allclose(served, engine, rtol=0, atol=1e-6)
