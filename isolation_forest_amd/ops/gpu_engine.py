@@ -343,6 +343,120 @@ def _eif_packed_wide(forest):
     return packed
 
 
+def _key32_inverse(k: np.ndarray) -> np.ndarray:
+    """Inverse of _key32 over the finite/inf key range (value of a key)."""
+    k = np.asarray(k, dtype=np.uint32)
+    pos = k >= np.uint32(0x80000000)
+    bits = np.where(pos, k - np.uint32(0x80000000),
+                    np.uint32(0x80000000) | (np.uint32(0x7FFFFFFF) - k))
+    return bits.astype(np.uint32).view(np.float32)
+
+
+def _eif0_eligible(forest, d: int) -> bool:
+    """extensionLevel-0 fast-path eligibility: every internal node has one
+    finite non-zero weight and a 12-bit coordinate (our builds give w=+-1;
+    foreign single-coordinate files qualify too)."""
+    flag = getattr(forest, "_eif0_ok", None)
+    if flag is None:
+        internal = forest.feature >= 0
+        w = forest.hyper_w[..., 0]
+        coord = forest.hyper_idx[..., 0]
+        flag = bool(
+            d <= 4094
+            and (not internal.any()
+                 or (np.isfinite(w[internal]).all()
+                     and (w[internal] != 0).all()
+                     and coord[internal].max(initial=0) <= 4094))
+        )
+        forest._eif0_ok = flag
+    return flag
+
+
+def _eif0_packed_v4(forest, d_sentinel: int, bf16: bool):
+    """Pack an extensionLevel-0 EIF forest into v4 node records with EXACT
+    integer-key thresholds, so scoring reuses the standard fixed-trip walk
+    (score_forest_v4<..., EIF0=true>) with NO per-visit multiply.
+
+    The oracle predicate per internal node is P(x) = f32(w*x) < offset32
+    (cpu_engine.path_lengths_extended, nnz=1). f32 multiplication by a
+    fixed finite non-zero w is monotone in x, so P is a threshold
+    predicate along the order-preserving key space: a vectorized binary
+    search (evaluating P with the same numpy f32 multiply the oracle
+    uses) finds, per node, the exact cut key K with
+        w > 0:  P(x) <=> key(x) <  K
+        w < 0:  P(x) <=> key(x) >= K   (flip bit 27; NaN guarded in-kernel)
+    The GPU performs no arithmetic on w at all, so the route is bitwise
+    vs the oracle BY CONSTRUCTION for every representable x."""
+    T, mn = forest.feature.shape
+    feat = forest.feature
+    internal = feat >= 0
+    leaf = feat == ExtendedForest.LEAF
+    depth = _node_depths(feat, forest.right)
+    w = forest.hyper_w[..., 0].astype(np.float32)
+    coord = forest.hyper_idx[..., 0].astype(np.int32)
+    o = forest.value.astype(np.float32)  # internal: offset32
+    flip = (w < 0) & internal
+
+    iw = w[internal]
+    io = o[internal]
+    iflip = flip[internal]
+
+    if bf16:
+        vals, keys = _bf16_key_tables()
+        lo = np.zeros(iw.shape, dtype=np.int64)
+        hi = np.full(iw.shape, len(vals), dtype=np.int64)
+        for _ in range(18):
+            mid = (lo + hi) // 2
+            midc = np.minimum(mid, len(vals) - 1)
+            pm = (iw * vals[midc]).astype(np.float32) < io
+            qm = np.where(iflip, ~pm, pm) & (mid < len(vals))
+            lo = np.where(qm, mid + 1, lo)
+            hi = np.where(qm, hi, mid)
+        k16 = np.where(hi < len(vals), keys[np.minimum(hi, len(vals) - 1)],
+                       np.int64(keys[-1]) + 1).astype(np.int64)
+        K = (k16 << 16).astype(np.uint64).astype(np.uint32)
+    else:
+        kmin = int(_key32(np.array([-np.inf], dtype=np.float32))[0])
+        kmax = int(_key32(np.array([np.inf], dtype=np.float32))[0])
+        lo = np.full(iw.shape, kmin, dtype=np.int64)
+        hi = np.full(iw.shape, kmax + 1, dtype=np.int64)
+        for _ in range(33):
+            mid = (lo + hi) // 2
+            xm = _key32_inverse(np.minimum(mid, kmax).astype(np.uint64))
+            pm = (iw * xm).astype(np.float32) < io
+            qm = np.where(iflip, ~pm, pm) & (mid <= kmax)
+            lo = np.where(qm, mid + 1, lo)
+            hi = np.where(qm, hi, mid)
+        K = hi.astype(np.uint64).astype(np.uint32)
+
+    ids = np.broadcast_to(np.arange(mn, dtype=np.int32)[None, :], (T, mn))
+    w0 = np.where(
+        internal,
+        coord | (forest.right.astype(np.int32) << 12)
+        | (flip.astype(np.int32) << 27),
+        np.int32(d_sentinel) | (ids << 12),
+    ).astype(np.int32)
+    w1 = np.zeros((T, mn), dtype=np.uint32)
+    w1[internal] = K
+    leafval = depth.astype(np.float32) + forest.value.astype(np.float32)
+    w1[leaf] = leafval[leaf].astype(np.float32).view(np.uint32)
+    pad = (-T) % 8
+    if pad:
+        w0p = np.empty((pad, mn), dtype=np.int32)
+        w0p[:] = np.int32(d_sentinel) | (ids[0] << 12)
+        w0 = np.concatenate([w0, w0p])
+        w1 = np.concatenate([w1, np.zeros((pad, mn), dtype=np.uint32)])
+    packed = np.empty((T + pad, mn, 2), dtype=np.int32)
+    packed[..., 0] = w0
+    packed[..., 1] = w1.view(np.int32)
+    ncount = np.concatenate(
+        [forest.node_count.astype(np.int32), np.ones(pad, dtype=np.int32)]
+    )
+    live = internal | leaf
+    max_depth = int(depth[live].max()) if live.any() else 0
+    return packed, ncount, max(max_depth, 1)
+
+
 def _densify_weights(hidx: np.ndarray, hw: np.ndarray, counts: np.ndarray,
                      D: int) -> np.ndarray:
     """Scatter sparse hyperplanes [T, mn, nnz] into dense rows [T, mn, D].
@@ -603,6 +717,12 @@ def _device_forest(model, device, v4_key=None):
                     hb[..., 0::2] | (hb[..., 1::2] << 16)
                 ).contiguous()
             extra["height"] = max_depth
+        elif isinstance(v4_key, tuple) and v4_key[0] == "eif0":
+            packed, ncount_np, max_depth = _eif0_packed_v4(
+                forest, v4_key[1], v4_key[2])
+            aos = torch.from_numpy(packed).to(device)
+            ncount = torch.from_numpy(ncount_np).to(device)
+            extra["height"] = max_depth
         elif isinstance(v4_key, tuple) and v4_key[0] == "wide":
             packed, max_depth = _nodes_packed_wide(forest, v4_key[1])
             aos = torch.from_numpy(packed).to(device)
@@ -712,6 +832,17 @@ def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torc
     # nnz <= 5 with uniform hyperplane widths -> fixed-trip sparse v2;
     # wider hyperplanes with d <= 32 -> densified dense v2 walk;
     # everything else -> the general strict-order kernel.
+    if (nnz == 1 and mn <= 32767 and _eif_uniform_nnz(forest)
+            and _eif0_eligible(forest, d)
+            and os.environ.get("IFA_EIF0_SPARSE") != "1"):
+        # extensionLevel-0: exact key-threshold transform -> the standard
+        # v4 walk (no per-visit multiply; bitwise by construction)
+        aos, ncount, extra = _device_forest(
+            model, X.device, v4_key=("eif0", d, X.dtype == torch.bfloat16))
+        return ext.score_forest_eif0(
+            X.contiguous(), aos, ncount, forest.num_trees, extra["height"],
+            c, finalize,
+        )
     if nnz <= 5 and _eif_uniform_nnz(forest):
         elem = 2 if X.dtype == torch.bfloat16 else 4
         dpad = d

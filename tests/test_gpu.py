@@ -598,6 +598,99 @@ class TestDeepForests:
             s_gpu.view(np.int32), s_cpu.view(np.int32))
 
 
+class TestEif0Route:
+    """extensionLevel-0 EIF routes through the v4 walk with exact key
+    thresholds (no per-visit multiply): bitwise vs the oracle, and
+    identical to the sparse v2 route it replaces."""
+
+    def _forest(self, seed=41, foreign=False):
+        rs = np.random.RandomState(seed)
+        X = rs.normal(size=(4000, 6)).astype(np.float32)
+        bag = cpu_engine.sample_bags(4000, 12, 256, seed=seed,
+                                     bootstrap=False)
+        fs = cpu_engine.feature_subsets(6, 6, 12, seed=seed)
+        forest = cpu_engine.build_extended_forest(X, bag, fs, seed, 256, 6,
+                                                  6, 0)
+        if foreign:
+            # foreign-style single-coordinate hyperplanes: arbitrary
+            # magnitudes/signs instead of our +-1-normalized weights
+            internal = forest.feature >= 0
+            w = rs.choice([0.5, -3.7, 1e-3, -1e4, 2.0],
+                          internal.sum()).astype(np.float32)
+            forest.hyper_w[..., 0][internal] = w
+            # plant knife-edge rows exactly at f32(o / w)
+            ik = np.argwhere(internal)
+            pick = ik[rs.randint(0, len(ik), size=400)]
+            for r, (t, n) in enumerate(pick):
+                with np.errstate(all="ignore"):
+                    X[r, forest.hyper_idx[t, n, 0]] = np.float32(
+                        forest.value[t, n] / forest.hyper_w[t, n, 0])
+        return forest, X
+
+    def _score(self, forest, X, dev, dtype=None):
+        from isolation_forest_amd.ops import gpu_engine
+
+        model = ExtendedIsolationForest(numEstimators=8,
+                                        extensionLevel=0).fit(X[:600])
+        model.forest = forest
+        model._gpu_forest_cache = {}
+        Xt = torch.from_numpy(X).to(dev)
+        if dtype is not None:
+            Xt = Xt.to(dtype)
+        return gpu_engine.score_extended_forest(model, Xt, finalize=False)
+
+    def test_f32_bitwise(self, dev):
+        forest, X = self._forest()
+        cpu_ps = cpu_engine.path_lengths_extended(forest, X)
+        gpu_ps = self._score(forest, X, dev)
+        np.testing.assert_array_equal(
+            gpu_ps.cpu().numpy().view(np.int32), cpu_ps.view(np.int32))
+
+    def test_bf16_bitwise(self, dev):
+        forest0, _ = self._forest(seed=42)
+        rs = np.random.RandomState(43)
+        Xb = torch.from_numpy(
+            rs.normal(size=(4000, 6)).astype(np.float32)).to(torch.bfloat16)
+        X = Xb.float().numpy()
+        cpu_ps = cpu_engine.path_lengths_extended(forest0, X)
+        from isolation_forest_amd.ops import gpu_engine
+
+        model = ExtendedIsolationForest(numEstimators=8,
+                                        extensionLevel=0).fit(X[:600])
+        model.forest = forest0
+        model._gpu_forest_cache = {}
+        gpu_ps = gpu_engine.score_extended_forest(
+            model, Xb.to(dev), finalize=False)
+        np.testing.assert_array_equal(
+            gpu_ps.cpu().numpy().view(np.int32), cpu_ps.view(np.int32))
+
+    def test_foreign_weights_knife_edge_bitwise(self, dev):
+        forest, X = self._forest(seed=44, foreign=True)
+        cpu_ps = cpu_engine.path_lengths_extended(forest, X)
+        gpu_ps = self._score(forest, X, dev)
+        np.testing.assert_array_equal(
+            gpu_ps.cpu().numpy().view(np.int32), cpu_ps.view(np.int32))
+
+    def test_matches_sparse_route(self, dev, monkeypatch):
+        forest, X = self._forest(seed=45)
+        s_eif0 = self._score(forest, X, dev).cpu().numpy()
+        monkeypatch.setenv("IFA_EIF0_SPARSE", "1")
+        s_sparse = self._score(forest, X, dev).cpu().numpy()
+        np.testing.assert_array_equal(
+            s_eif0.view(np.int32), s_sparse.view(np.int32))
+
+    def test_end_to_end_ext0_fit(self, dev):
+        rs = np.random.RandomState(46)
+        X = torch.from_numpy(
+            rs.normal(size=(20000, 9)).astype(np.float32)).to(dev)
+        model = ExtendedIsolationForest(
+            numEstimators=100, extensionLevel=0, randomSeed=2).fit(X)
+        s_gpu = model.score(X).cpu().numpy()
+        s_cpu = model.score(X.float().cpu()).numpy()
+        np.testing.assert_array_equal(
+            s_gpu.view(np.int32), s_cpu.view(np.int32))
+
+
 class TestForeignF64SplitsGPU:
     def test_knife_edge_foreign_model_bitwise(self, dev):
         """Foreign f64 splits strictly between adjacent f32s, rows planted

@@ -263,3 +263,82 @@ class TestWidePacking:
         np.testing.assert_array_equal(
             wide[..., 2].view(np.float32)[internal],
             forest.value.astype(np.float32)[internal])
+
+
+class TestEif0KeyThresholds:
+    """extensionLevel-0 pack: the binary-searched key thresholds must
+    reproduce P(x) = f32(w*x) < o EXACTLY for every x (CPU check of the
+    same table the GPU walk consumes)."""
+
+    def _check(self, bf16, n_nodes=500, seed=0):
+        from isolation_forest_amd.core.forest import ExtendedForest
+        from isolation_forest_amd.ops.gpu_engine import (
+            _bf16_key_tables, _eif0_packed_v4, _key32)
+
+        rs = np.random.RandomState(seed)
+        # synthetic single-node-per-tree forests: one internal + two leaves
+        T, mn = n_nodes, 3
+        f = ExtendedForest(
+            feature=np.zeros((T, mn), dtype=np.int32),
+            value=np.zeros((T, mn), dtype=np.float32),
+            right=np.zeros((T, mn), dtype=np.int32),
+            num_instances=np.full((T, mn), 1, dtype=np.int64),
+            node_count=np.full(T, 3, dtype=np.int32),
+            hyper_idx=np.zeros((T, mn, 1), dtype=np.int32),
+            hyper_w=np.zeros((T, mn, 1), dtype=np.float32),
+            offset64=np.zeros((T, mn), dtype=np.float64),
+            num_samples=2, num_features=4, total_num_features=4,
+            extension_level=0,
+        )
+        f.feature[:, 0] = 1   # internal root, nnz count = 1
+        f.feature[:, 1:] = -1
+        f.right[:, 0] = 2
+        f.right[:, 1] = 1
+        f.right[:, 2] = 2
+        w = rs.choice([1.0, -1.0, 0.5, -3.7, 1e-3, -1e20, 1e20],
+                      T).astype(np.float32)
+        w *= rs.uniform(0.5, 2.0, T).astype(np.float32)
+        o = (rs.standard_normal(T) * rs.choice([1e-4, 1.0, 1e4], T)
+             ).astype(np.float32)
+        f.hyper_w[:, 0, 0] = w
+        f.value[:, 0] = o
+        packed, _, _ = _eif0_packed_v4(f, 4, bf16)
+        K = packed[:, 0, 1].view(np.uint32)
+        flip = (packed[:, 0, 0] >> 27) & 1
+
+        # probe values: random + knife edges around o/w + specials
+        probes = [rs.standard_normal(64).astype(np.float32) * s
+                  for s in (1e-6, 1.0, 1e6)]
+        with np.errstate(all="ignore"):
+            cut = (o / w).astype(np.float32)
+        near = [np.nextafter(cut, np.float32(np.inf)),
+                np.nextafter(cut, np.float32(-np.inf)), cut]
+        specials = np.array([0.0, -0.0, np.inf, -np.inf, 1e-45, -1e-45],
+                            dtype=np.float32)
+        for t in range(0, T, 7):
+            xs = np.concatenate(
+                [p for p in probes] + [np.array([n[t] for n in near],
+                                                dtype=np.float32), specials])
+            if bf16:
+                import torch as _t
+
+                xs = _t.from_numpy(xs).to(_t.bfloat16).float().numpy()
+                vals, keys = _bf16_key_tables()
+                idx = np.searchsorted(vals, xs)
+                k_row = (keys[np.minimum(idx, len(keys) - 1)].astype(
+                    np.uint64) << 16 | 0xFFFF).astype(np.uint64)
+            else:
+                k_row = _key32(xs).astype(np.uint64)
+            P = (np.float32(w[t]) * xs).astype(np.float32) < o[t]
+            if flip[t]:
+                got = (k_row >= np.uint64(K[t])) & (k_row != 0xFFFFFFFF)
+            else:
+                got = k_row < np.uint64(K[t])
+            np.testing.assert_array_equal(got, P, err_msg=f"node {t} "
+                                          f"w={w[t]} o={o[t]} bf16={bf16}")
+
+    def test_f32_exact(self):
+        self._check(bf16=False)
+
+    def test_bf16_exact(self):
+        self._check(bf16=True)
