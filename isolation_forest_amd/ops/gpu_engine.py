@@ -373,20 +373,34 @@ def _eif0_eligible(forest, d: int) -> bool:
 
 
 def _eif0_packed_v4(forest, d_sentinel: int, bf16: bool):
-    """Pack an extensionLevel-0 EIF forest into v4 node records with EXACT
-    integer-key thresholds, so scoring reuses the standard fixed-trip walk
-    (score_forest_v4<..., EIF0=true>) with NO per-visit multiply.
+    """Pack an extensionLevel-0 EIF forest into STANDARD v4 node records,
+    so scoring reuses the unmodified fixed-trip walk (score_forest_v4)
+    with NO per-visit multiply and NO extra compare.
 
-    The oracle predicate per internal node is P(x) = f32(w*x) < offset32
-    (cpu_engine.path_lengths_extended, nnz=1). f32 multiplication by a
-    fixed finite non-zero w is monotone in x, so P is a threshold
-    predicate along the order-preserving key space: a vectorized binary
-    search (evaluating P with the same numpy f32 multiply the oracle
-    uses) finds, per node, the exact cut key K with
-        w > 0:  P(x) <=> key(x) <  K
-        w < 0:  P(x) <=> key(x) >= K   (flip bit 27; NaN guarded in-kernel)
+    Two exact transformations compose:
+
+    1. Thresholds. The oracle predicate per internal node is
+       P(x) = f32(w*x) < offset32 (cpu_engine.path_lengths_extended,
+       nnz=1). f32 multiplication by a fixed finite non-zero w is
+       monotone in x, so P is a threshold predicate along the
+       order-preserving key space: a vectorized binary search (evaluating
+       P with the same numpy f32 multiply the oracle uses) finds, per
+       node, the exact cut key K with
+           w > 0:  P(x) <=> key(x) <  K
+           w < 0:  P(x) <=> key(x) >= K
+    2. Subtree mirroring. Negative-weight nodes (w < 0) have their left
+       and right SUBTREES swapped in a re-materialized pre-order layout,
+       turning "go left iff key >= K" into the v4 kernel's native
+       "go left iff key < K" (the compare-false branch is the old left).
+       Node depths — the only structural quantity scoring uses — are
+       invariant under mirroring.
+
     The GPU performs no arithmetic on w at all, so the route is bitwise
-    vs the oracle BY CONSTRUCTION for every representable x."""
+    vs the oracle BY CONSTRUCTION for every representable non-NaN x.
+    NaN features are the one exception (the oracle's NaN dot compares
+    false at EVERY node, which no single mirrored threshold can encode):
+    callers route rows containing NaN to the strict-order kernels
+    (score_extended_forest checks torch.isnan once per call)."""
     T, mn = forest.feature.shape
     feat = forest.feature
     internal = feat >= 0
@@ -429,17 +443,51 @@ def _eif0_packed_v4(forest, d_sentinel: int, bf16: bool):
             hi = np.where(qm, hi, mid)
         K = hi.astype(np.uint64).astype(np.uint32)
 
+    # --- mirrored pre-order re-layout (level-synchronous, vectorized) ---
+    right = forest.right.astype(np.int64)
+    nc = forest.node_count.astype(np.int64)
+    newpos = np.zeros((T, mn), dtype=np.int64)
+    end_old = np.zeros((T, mn), dtype=np.int64)  # old-coords subtree end
+    end_old[np.arange(T), 0] = nc
+    new_rightpos = np.zeros((T, mn), dtype=np.int64)
+    ft = np.arange(T, dtype=np.int64)
+    fn = np.zeros(T, dtype=np.int64)
+    while len(ft):
+        isint = internal[ft, fn]
+        ft, fn = ft[isint], fn[isint]
+        if not len(ft):
+            break
+        rt = right[ft, fn]
+        s_left = rt - (fn + 1)
+        s_right = end_old[ft, fn] - rt
+        fl = flip[ft, fn]
+        lpos = newpos[ft, fn] + 1
+        # new left child = old right when flipped (compare-true branch)
+        newpos[ft, fn + 1] = np.where(fl, lpos + s_right, lpos)
+        newpos[ft, rt] = np.where(fl, lpos, lpos + s_left)
+        new_rightpos[ft, fn] = np.where(fl, newpos[ft, fn + 1],
+                                        newpos[ft, rt])
+        end_old[ft, fn + 1] = rt
+        end_old[ft, rt] = end_old[ft, fn]
+        ft = np.concatenate([ft, ft])
+        fn = np.concatenate([fn + 1, rt])
+
     ids = np.broadcast_to(np.arange(mn, dtype=np.int32)[None, :], (T, mn))
-    w0 = np.where(
-        internal,
-        coord | (forest.right.astype(np.int32) << 12)
-        | (flip.astype(np.int32) << 27),
-        np.int32(d_sentinel) | (ids << 12),
-    ).astype(np.int32)
+    w0 = np.empty((T, mn), dtype=np.int32)
+    w0[:] = np.int32(d_sentinel) | (ids << 12)  # pad slots: self-loop
     w1 = np.zeros((T, mn), dtype=np.uint32)
-    w1[internal] = K
-    leafval = depth.astype(np.float32) + forest.value.astype(np.float32)
-    w1[leaf] = leafval[leaf].astype(np.float32).view(np.uint32)
+    ti = np.broadcast_to(np.arange(T, dtype=np.int64)[:, None], (T, mn))
+    # scatter live nodes to their mirrored positions
+    it_, in_ = ti[internal], newpos[internal]
+    w0[it_, in_] = (coord[internal]
+                    | (new_rightpos[internal].astype(np.int32) << 12))
+    w1[it_, in_] = K
+    lt_, ln_ = ti[leaf], newpos[leaf]
+    w0[lt_, ln_] = np.int32(d_sentinel) | (ln_.astype(np.int32) << 12)
+    leafval = (depth.astype(np.float32)
+               + forest.value.astype(np.float32))[leaf]
+    w1[lt_, ln_] = leafval.astype(np.float32).view(np.uint32)
+
     pad = (-T) % 8
     if pad:
         w0p = np.empty((pad, mn), dtype=np.int32)
@@ -834,12 +882,16 @@ def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torc
     # everything else -> the general strict-order kernel.
     if (nnz == 1 and mn <= 32767 and _eif_uniform_nnz(forest)
             and _eif0_eligible(forest, d)
-            and os.environ.get("IFA_EIF0_SPARSE") != "1"):
-        # extensionLevel-0: exact key-threshold transform -> the standard
-        # v4 walk (no per-visit multiply; bitwise by construction)
+            and os.environ.get("IFA_EIF0_SPARSE") != "1"
+            and not bool(torch.isnan(X).any().item())):
+        # extensionLevel-0: exact key thresholds + mirrored subtrees ->
+        # the UNMODIFIED standard v4 walk (no per-visit multiply; bitwise
+        # by construction for non-NaN rows — NaN rows take the
+        # strict-order route below, whose dot reproduces the oracle's
+        # NaN-compares-false behaviour)
         aos, ncount, extra = _device_forest(
             model, X.device, v4_key=("eif0", d, X.dtype == torch.bfloat16))
-        return ext.score_forest_eif0(
+        return ext.score_forest(
             X.contiguous(), aos, ncount, forest.num_trees, extra["height"],
             c, finalize,
         )

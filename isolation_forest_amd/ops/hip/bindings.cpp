@@ -34,7 +34,7 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
                                   size_t lds, hipStream_t stream);
 
 void launch_score_forest(bool bf16, int rpt, bool rows_lds, bool nodes_lds,
-                         int ilp, int eif0, const void* X, const void* nodes,
+                         int ilp, const void* X, const void* nodes,
                          const int32_t* ncount, float* out, int64_t N,
                          int32_t d, int32_t dpad, int32_t Tpad,
                          int32_t max_nodes, int32_t height_limit, float fT,
@@ -227,10 +227,10 @@ std::vector<torch::Tensor> build_extended_forest(
   return {feat, value, right, count, ncount, hidx, hw, off64, depth};
 }
 
-torch::Tensor score_forest_impl(torch::Tensor X, torch::Tensor nodes_packed,
-                                torch::Tensor ncount, int64_t num_trees,
-                                int64_t height_limit, double c_norm,
-                                bool finalize, bool eif0) {
+torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
+                           torch::Tensor ncount, int64_t num_trees,
+                           int64_t height_limit, double c_norm,
+                           bool finalize) {
   CHECK_CUDA(X);
   CHECK_CONTIG(X);
   CHECK_CUDA(nodes_packed);
@@ -289,8 +289,7 @@ torch::Tensor score_forest_impl(torch::Tensor X, torch::Tensor nodes_packed,
   int64_t rows_per_block = (int64_t)(rows_lds ? rpt : 1) * 256;
   int blocks = (int)std::min<int64_t>(
       (N + rows_per_block - 1) / rows_per_block, 8192);
-  ifa::launch_score_forest(bf16, rpt, rows_lds, nodes_lds, ilp,
-                           eif0 ? 1 : 0, X.data_ptr(),
+  ifa::launch_score_forest(bf16, rpt, rows_lds, nodes_lds, ilp, X.data_ptr(),
                            nodes_packed.data_ptr<int32_t>(),
                            ncount.data_ptr<int32_t>(), out.data_ptr<float>(),
                            N, (int32_t)d, (int32_t)dpad, (int32_t)Tpad,
@@ -300,22 +299,6 @@ torch::Tensor score_forest_impl(torch::Tensor X, torch::Tensor nodes_packed,
   return out;
 }
 
-torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
-                           torch::Tensor ncount, int64_t num_trees,
-                           int64_t height_limit, double c_norm,
-                           bool finalize) {
-  return score_forest_impl(X, nodes_packed, ncount, num_trees, height_limit,
-                           c_norm, finalize, false);
-}
-
-torch::Tensor score_forest_eif0(torch::Tensor X, torch::Tensor nodes_packed,
-                                torch::Tensor ncount, int64_t num_trees,
-                                int64_t height_limit, double c_norm,
-                                bool finalize) {
-  // extensionLevel-0 EIF packed as exact key thresholds (flip bit 27)
-  return score_forest_impl(X, nodes_packed, ncount, num_trees, height_limit,
-                           c_norm, finalize, true);
-}
 
 torch::Tensor score_extended_forest(torch::Tensor X,
                                     torch::Tensor nodes_packed,
@@ -580,9 +563,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("build_extended_forest", &build_extended_forest,
         "build extended iTrees (K3-K5)");
   m.def("score_forest", &score_forest, "batched path-length scoring (K6)");
-  m.def("score_forest_eif0", &score_forest_eif0,
-        "extensionLevel-0 EIF scoring through the v4 walk (exact key "
-        "thresholds, flip-aware compare)");
   m.def("score_extended_forest", &score_extended_forest,
         "batched EIF scoring (K7)");
   m.def("score_extended_sparse_v2", &score_extended_sparse_v2,

@@ -529,8 +529,7 @@ __device__ __host__ __forceinline__ int row_stride(int d) {
 }
 
 
-template <typename KT, int RPT, bool ROWS_LDS, int V4_ILP, bool NODES_LDS,
-          bool EIF0 = false>
+template <typename KT, int RPT, bool ROWS_LDS, int V4_ILP, bool NODES_LDS>
 __global__ void __launch_bounds__(256) score_forest_v4(
     const KT* __restrict__ X,          // raw bf16/f32 bits [N][d]
     const int2* __restrict__ nodes,    // [Tpad][max_nodes] packed v4
@@ -631,19 +630,8 @@ __global__ void __launch_bounds__(256) score_forest_v4(
               x = (f < d) ? widen_key(key_of_bits<KT>(kraw[r][s]))
                           : 0xFFFFFFFFu;
             }
-            bool left;
-            if (EIF0) {
-              const uint32_t K = (uint32_t)nd[r][s].y;
-              // flip (bit 27) = negative single weight: go left iff
-              // x > cut, i.e. key >= K — except the NaN key (dot NaN
-              // compares false in the oracle, so NaN always routes right)
-              left = (nd[r][s].x >> 27) & 1
-                         ? (x >= K && x != 0xFFFFFFFFu)
-                         : (x < K);
-            } else {
-              left = x < (uint32_t)nd[r][s].y;
-            }
-            cur[r][s] = left ? cur[r][s] + 1 : pn_right(nd[r][s].x);
+            cur[r][s] = (x < (uint32_t)nd[r][s].y) ? cur[r][s] + 1
+                                                   : pn_right(nd[r][s].x);
           }
         }
       }
@@ -1471,29 +1459,26 @@ void launch_build_extended_forest(const float* bags, const int32_t* feat_sub,
 }
 
 void launch_score_forest(bool bf16, int rpt, bool rows_lds, bool nodes_lds,
-                         int ilp, int eif0, const void* X, const void* nodes,
+                         int ilp, const void* X, const void* nodes,
                          const int32_t* ncount, float* out, int64_t N,
                          int32_t d, int32_t dpad, int32_t Tpad,
                          int32_t max_nodes, int32_t height_limit, float fT,
                          float c_norm, int finalize, size_t lds, int blocks,
                          hipStream_t stream) {
-#define LS1(KT, RPT, RL, TI, NL, E0)                                          \
-  do {                                                                        \
-    raise_lds((const void*)score_forest_v4<KT, RPT, RL, TI, NL, E0>, lds);    \
-    hipLaunchKernelGGL((score_forest_v4<KT, RPT, RL, TI, NL, E0>),            \
-                       dim3(blocks), dim3(256), lds, stream, (const KT*)X,    \
-                       (const int2*)nodes, ncount, out, N, d, dpad, Tpad,     \
-                       max_nodes, height_limit, fT, c_norm, finalize);        \
-  } while (0)
 #define LS(KT, RPT, RL, TI)                                                   \
   do {                                                                        \
-    if (eif0) {  /* extensionLevel-0 route: ILP fixed at 4 */                 \
-      if (nodes_lds) LS1(KT, RPT, RL, 4, true, true);                         \
-      else LS1(KT, RPT, RL, 4, false, true);                                  \
-    } else if (nodes_lds) {                                                   \
-      LS1(KT, RPT, RL, TI, true, false);                                      \
+    if (nodes_lds) {                                                          \
+      raise_lds((const void*)score_forest_v4<KT, RPT, RL, TI, true>, lds);    \
+      hipLaunchKernelGGL((score_forest_v4<KT, RPT, RL, TI, true>),            \
+                         dim3(blocks), dim3(256), lds, stream, (const KT*)X,  \
+                         (const int2*)nodes, ncount, out, N, d, dpad, Tpad,   \
+                         max_nodes, height_limit, fT, c_norm, finalize);      \
     } else {                                                                  \
-      LS1(KT, RPT, RL, TI, false, false);                                     \
+      raise_lds((const void*)score_forest_v4<KT, RPT, RL, TI, false>, lds);   \
+      hipLaunchKernelGGL((score_forest_v4<KT, RPT, RL, TI, false>),           \
+                         dim3(blocks), dim3(256), lds, stream, (const KT*)X,  \
+                         (const int2*)nodes, ncount, out, N, d, dpad, Tpad,   \
+                         max_nodes, height_limit, fT, c_norm, finalize);      \
     }                                                                         \
   } while (0)
 #define LS_ILP(KT, RPT, RL)                                                   \
@@ -1511,7 +1496,6 @@ void launch_score_forest(bool bf16, int rpt, bool rows_lds, bool nodes_lds,
   }
 #undef LS_ILP
 #undef LS
-#undef LS1
 }
 
 void launch_score_extended_dense(bool bf16, bool rows_lds, bool wlds,

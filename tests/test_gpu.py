@@ -679,6 +679,18 @@ class TestEif0Route:
         np.testing.assert_array_equal(
             s_eif0.view(np.int32), s_sparse.view(np.int32))
 
+    def test_nan_rows_route_to_strict_kernel(self, dev):
+        """NaN features cannot take the mirrored-threshold walk (the
+        oracle's NaN dot compares false at EVERY node); the route must
+        detect them and stay bitwise via the strict-order kernel."""
+        forest, X = self._forest(seed=47)
+        rs = np.random.RandomState(48)
+        X[rs.randint(0, len(X), 50), rs.randint(0, 6, 50)] = np.nan
+        cpu_ps = cpu_engine.path_lengths_extended(forest, X)
+        gpu_ps = self._score(forest, X, dev)
+        np.testing.assert_array_equal(
+            gpu_ps.cpu().numpy().view(np.int32), cpu_ps.view(np.int32))
+
     def test_end_to_end_ext0_fit(self, dev):
         rs = np.random.RandomState(46)
         X = torch.from_numpy(

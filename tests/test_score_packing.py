@@ -265,80 +265,105 @@ class TestWidePacking:
             forest.value.astype(np.float32)[internal])
 
 
-class TestEif0KeyThresholds:
-    """extensionLevel-0 pack: the binary-searched key thresholds must
-    reproduce P(x) = f32(w*x) < o EXACTLY for every x (CPU check of the
-    same table the GPU walk consumes)."""
+class TestEif0MirrorPack:
+    """extensionLevel-0 pack: exact key thresholds + mirrored subtrees
+    must make the UNMODIFIED v4 walk reproduce the oracle bitwise. The
+    v4 walk is simulated in numpy here (same simulation proven against
+    the GPU kernel in TestPackedV4); GPU execution is covered by
+    tests/test_gpu.py::TestEif0Route."""
 
-    def _check(self, bf16, n_nodes=500, seed=0):
-        from isolation_forest_amd.core.forest import ExtendedForest
+    def _simulate_and_compare(self, forest, X, bf16):
+        import torch as _t
+
         from isolation_forest_amd.ops.gpu_engine import (
             _bf16_key_tables, _eif0_packed_v4, _key32)
 
+        if bf16:
+            X = _t.from_numpy(X).to(_t.bfloat16).float().numpy()
+        oracle = cpu_engine.path_lengths_extended(forest, X)
+        d = X.shape[1]
+        packed, ncount, max_depth = _eif0_packed_v4(forest, d, bf16)
+        w0 = packed[:, :, 0]
+        w1 = packed[:, :, 1].view(np.uint32)
+        if bf16:
+            vals, keys = _bf16_key_tables()
+            idx = np.searchsorted(vals, X.reshape(-1))
+            k = keys[np.minimum(idx, len(keys) - 1)].astype(np.uint32)
+            keys_m = ((k.astype(np.uint64) << 16) | 0xFFFF).astype(
+                np.uint64).reshape(X.shape)
+        else:
+            keys_m = _key32(X.reshape(-1)).astype(np.uint64).reshape(X.shape)
+        keys_m = np.concatenate(
+            [keys_m, np.full((X.shape[0], 1), 0xFFFFFFFF,
+                             dtype=np.uint64)], axis=1)
+        total = np.zeros(X.shape[0], dtype=np.float32)
+        for t in range(forest.num_trees):
+            cur = np.zeros(X.shape[0], dtype=np.int64)
+            for _ in range(max_depth):
+                f = w0[t, cur] & 0xFFF
+                right = (w0[t, cur] >> 12) & 0x7FFF
+                x = keys_m[np.arange(X.shape[0]), f]
+                cur = np.where(x < w1[t, cur].astype(np.uint64),
+                               cur + 1, right)
+            total = (total + w1[t, cur].view(np.float32)).astype(np.float32)
+        np.testing.assert_array_equal(total.view(np.int32),
+                                      oracle.view(np.int32))
+
+    def _ext0_forest(self, seed, foreign=False, knife=False):
         rs = np.random.RandomState(seed)
-        # synthetic single-node-per-tree forests: one internal + two leaves
-        T, mn = n_nodes, 3
-        f = ExtendedForest(
-            feature=np.zeros((T, mn), dtype=np.int32),
-            value=np.zeros((T, mn), dtype=np.float32),
-            right=np.zeros((T, mn), dtype=np.int32),
-            num_instances=np.full((T, mn), 1, dtype=np.int64),
-            node_count=np.full(T, 3, dtype=np.int32),
-            hyper_idx=np.zeros((T, mn, 1), dtype=np.int32),
-            hyper_w=np.zeros((T, mn, 1), dtype=np.float32),
-            offset64=np.zeros((T, mn), dtype=np.float64),
-            num_samples=2, num_features=4, total_num_features=4,
-            extension_level=0,
-        )
-        f.feature[:, 0] = 1   # internal root, nnz count = 1
-        f.feature[:, 1:] = -1
-        f.right[:, 0] = 2
-        f.right[:, 1] = 1
-        f.right[:, 2] = 2
-        w = rs.choice([1.0, -1.0, 0.5, -3.7, 1e-3, -1e20, 1e20],
-                      T).astype(np.float32)
-        w *= rs.uniform(0.5, 2.0, T).astype(np.float32)
-        o = (rs.standard_normal(T) * rs.choice([1e-4, 1.0, 1e4], T)
-             ).astype(np.float32)
-        f.hyper_w[:, 0, 0] = w
-        f.value[:, 0] = o
-        packed, _, _ = _eif0_packed_v4(f, 4, bf16)
-        K = packed[:, 0, 1].view(np.uint32)
-        flip = (packed[:, 0, 0] >> 27) & 1
+        X = rs.normal(size=(2000, 5)).astype(np.float32)
+        bag = cpu_engine.sample_bags(2000, 8, 128, seed=seed,
+                                     bootstrap=False)
+        fs = cpu_engine.feature_subsets(5, 5, 8, seed=seed)
+        forest = cpu_engine.build_extended_forest(X, bag, fs, seed, 128, 5,
+                                                  5, 0)
+        internal = forest.feature >= 0
+        if foreign:
+            w = rs.choice([0.5, -3.7, 1e-3, -1e4, 2.0, -1e-20],
+                          internal.sum()).astype(np.float32)
+            forest.hyper_w[..., 0][internal] = w
+        if knife:
+            ik = np.argwhere(internal)
+            pick = ik[rs.randint(0, len(ik), size=300)]
+            with np.errstate(all="ignore"):
+                for r, (t, n) in enumerate(pick):
+                    X[r, forest.hyper_idx[t, n, 0]] = np.float32(
+                        forest.value[t, n] / forest.hyper_w[t, n, 0])
+        return forest, X
 
-        # probe values: random + knife edges around o/w + specials
-        probes = [rs.standard_normal(64).astype(np.float32) * s
-                  for s in (1e-6, 1.0, 1e6)]
-        with np.errstate(all="ignore"):
-            cut = (o / w).astype(np.float32)
-        near = [np.nextafter(cut, np.float32(np.inf)),
-                np.nextafter(cut, np.float32(-np.inf)), cut]
-        specials = np.array([0.0, -0.0, np.inf, -np.inf, 1e-45, -1e-45],
-                            dtype=np.float32)
-        for t in range(0, T, 7):
-            xs = np.concatenate(
-                [p for p in probes] + [np.array([n[t] for n in near],
-                                                dtype=np.float32), specials])
-            if bf16:
-                import torch as _t
+    def test_native_weights_f32(self):
+        forest, X = self._ext0_forest(1)
+        self._simulate_and_compare(forest, X, bf16=False)
 
-                xs = _t.from_numpy(xs).to(_t.bfloat16).float().numpy()
-                vals, keys = _bf16_key_tables()
-                idx = np.searchsorted(vals, xs)
-                k_row = (keys[np.minimum(idx, len(keys) - 1)].astype(
-                    np.uint64) << 16 | 0xFFFF).astype(np.uint64)
-            else:
-                k_row = _key32(xs).astype(np.uint64)
-            P = (np.float32(w[t]) * xs).astype(np.float32) < o[t]
-            if flip[t]:
-                got = (k_row >= np.uint64(K[t])) & (k_row != 0xFFFFFFFF)
-            else:
-                got = k_row < np.uint64(K[t])
-            np.testing.assert_array_equal(got, P, err_msg=f"node {t} "
-                                          f"w={w[t]} o={o[t]} bf16={bf16}")
+    def test_native_weights_bf16(self):
+        forest, X = self._ext0_forest(2)
+        self._simulate_and_compare(forest, X, bf16=True)
 
-    def test_f32_exact(self):
-        self._check(bf16=False)
+    def test_foreign_weights_knife_edge_f32(self):
+        forest, X = self._ext0_forest(3, foreign=True, knife=True)
+        self._simulate_and_compare(forest, X, bf16=False)
 
-    def test_bf16_exact(self):
-        self._check(bf16=True)
+    def test_foreign_weights_bf16(self):
+        forest, X = self._ext0_forest(4, foreign=True)
+        self._simulate_and_compare(forest, X, bf16=True)
+
+    def test_specials(self):
+        """zeros, denormals and infs in the data route exactly."""
+        forest, X = self._ext0_forest(5, foreign=True)
+        rs = np.random.RandomState(6)
+        sp = np.array([0.0, -0.0, np.inf, -np.inf, 1e-45, -1e-45],
+                      dtype=np.float32)
+        X[rs.randint(0, len(X), 200), rs.randint(0, 5, 200)] = rs.choice(
+            sp, 200)
+        self._simulate_and_compare(forest, X, bf16=False)
+
+    def test_mirror_preserves_depths(self):
+        from isolation_forest_amd.ops.gpu_engine import (
+            _eif0_packed_v4, _node_depths)
+
+        forest, X = self._ext0_forest(7, foreign=True)
+        packed, ncount, max_depth = _eif0_packed_v4(forest, 5, False)
+        depth = _node_depths(forest.feature, forest.right)
+        live = forest.feature >= -1
+        assert max_depth == int(depth[(forest.feature >= 0)
+                                      | (forest.feature == -1)].max())

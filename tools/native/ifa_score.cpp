@@ -43,7 +43,7 @@
 
 namespace ifa {
 void launch_score_forest(bool bf16, int rpt, bool rows_lds, bool nodes_lds,
-                         int ilp, int eif0, const void* X, const void* nodes,
+                         int ilp, const void* X, const void* nodes,
                          const int32_t* ncount, float* out, int64_t N,
                          int32_t d, int32_t dpad, int32_t Tpad,
                          int32_t max_nodes, int32_t height_limit, float fT,
@@ -632,7 +632,7 @@ int main(int argc, char** argv) {
     int64_t rows_per_block = (int64_t)(rows_lds ? rpt : 1) * 256;
     int blocks = (int)std::min<int64_t>(
         (N + rows_per_block - 1) / rows_per_block, 8192);
-    ifa::launch_score_forest(bf16, rpt, rows_lds, nodes_lds, 4, 0, dX, dNodes,
+    ifa::launch_score_forest(bf16, rpt, rows_lds, nodes_lds, 4, dX, dNodes,
                              (const int32_t*)dNcount, (float*)dOut, N, d,
                              (int32_t)dpad, p.Tpad, p.mn, p.max_depth,
                              (float)p.T, c_norm, 1, lds, blocks, 0);
