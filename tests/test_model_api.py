@@ -335,3 +335,47 @@ class TestSklearnAdapter:
         ])
         labels = pipe.fit_predict(X)
         assert labels.shape == (len(X),)
+
+
+class TestExactQuantilePathological:
+    """adversarially concentrated score distributions
+    must not break the exact histogram-refinement quantile (bounded
+    passes + comparator-anchored ranks + exact finish)."""
+
+    def test_single_ulp_cluster(self):
+        from isolation_forest_amd.core import threshold as th
+
+        # 2M scores inside one float32 ulp neighbourhood around 0.5, plus
+        # a handful of outliers: every histogram pass sees ties
+        base = np.float32(0.5)
+        vals = np.array(
+            [base, np.nextafter(base, np.float32(1.0)),
+             np.nextafter(base, np.float32(0.0))], dtype=np.float32)
+        rs = np.random.RandomState(0)
+        s = vals[rs.randint(0, 3, size=2_000_000)]
+        s[:5] = np.float32(0.9)
+        t = torch.from_numpy(s)
+        for contamination in (0.3, 1e-6):
+            got = th.compute_threshold(t, contamination, 0.0)
+            k = max(1, min(len(s), int(np.ceil((1 - contamination) * len(s)))))
+            expect = float(np.sort(s)[k - 1])
+            assert got == expect, (contamination, got, expect)
+
+    def test_all_identical(self):
+        from isolation_forest_amd.core import threshold as th
+
+        s = torch.full((100_000,), 0.47)
+        assert th.compute_threshold(s, 0.1, 0.0) == pytest.approx(0.47)
+
+    def test_two_spikes_far_apart(self):
+        from isolation_forest_amd.core import threshold as th
+
+        rs = np.random.RandomState(1)
+        s = np.where(rs.rand(1_000_000) < 0.999,
+                     np.float32(1e-30), np.float32(1e30)).astype(np.float32)
+        t = torch.from_numpy(s)
+        for contamination in (0.0005, 0.002, 0.3):
+            got = th.compute_threshold(t, contamination, 0.0)
+            k = max(1, min(len(s), int(np.ceil((1 - contamination) * len(s)))))
+            expect = float(np.sort(s)[k - 1])
+            assert got == expect, contamination
