@@ -192,3 +192,33 @@ class TestExtendedSharded:
         res = _run_workers(_body_extended_sharded)
         assert res[0]["fp"] == res[1]["fp"]
         assert res[0]["ext"] == 3
+
+
+class TestWorldFour:
+    """The reference's distributed tests run on a real local[4] session
+    (core/TestUtils.scala:39-56); world-4 gloo is our analog of that
+    parallelism degree — same collectives, uneven 11/4 tree shards."""
+
+    def test_tree_sharded_world4_bitwise(self):
+        res = _run_workers(_body_tree_sharded_fit, world=4)
+        for r in range(1, 4):
+            assert res[0]["fp"] == res[r]["fp"]
+            assert res[0]["threshold"] == res[r]["threshold"]
+        from isolation_forest_amd import IsolationForest
+
+        rs = np.random.RandomState(0)
+        X = rs.normal(size=(3000, 5)).astype(np.float32)
+        solo = IsolationForest(numEstimators=11, randomSeed=5,
+                               contamination=0.05,
+                               contaminationError=0.0).fit(X)
+        assert res[0]["fp"] == [solo.forest.tree_to_string(t)
+                                for t in range(11)]
+        assert res[0]["threshold"] == solo.outlier_score_threshold
+
+    def test_row_sharded_world4(self):
+        res = _run_workers(_body_row_sharded_fit, world=4)
+        for r in range(4):
+            assert res[r]["auroc"] > 0.85
+            assert res[r]["trees"] == 40
+            assert res[r]["observed"] == pytest.approx(0.02, abs=0.01)
+        assert len({res[r]["threshold"] for r in range(4)}) == 1
