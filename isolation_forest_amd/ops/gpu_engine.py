@@ -913,8 +913,17 @@ def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torc
                 X.contiguous(), aos, extra["values"], extra["hidx"],
                 extra["hw"], ncount, extra["height"], c, finalize,
             )
-    if d <= 32 and (nnz == d or nnz >= 6):
-        D = 8 if d <= 8 else (16 if d <= 16 else 32)
+    if d <= 128 and (nnz == d or nnz >= 6):
+        if d <= 8:
+            D = 8
+        elif d <= 16:
+            D = 16
+        elif d <= 32:
+            D = 32
+        elif d <= 64:
+            D = 64
+        else:
+            D = 128
         mn = forest.feature.shape[1]
         use_v3 = (X.dtype == torch.bfloat16
                   and os.environ.get("IFA_EIF_DENSE_V2") != "1")
@@ -927,14 +936,15 @@ def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torc
                     X.contiguous(), aos, extra["values"], extra["hwp"],
                     ncount, extra["height"], c, finalize,
                 )
-        lds = mn * 12 + 16 + mn * (D // 4 + 1) * 16
-        if lds <= 160 * 1024:
-            aos, ncount, extra = _device_forest(
-                model, X.device, v4_key=("eif_dense", D))
-            return ext.score_extended_dense_v2(
-                X.contiguous(), aos, extra["values"], extra["hw"], ncount,
-                extra["height"], c, finalize,
-            )
+        if d <= 64:  # v2's f32 staging exceeds LDS past D=64
+            lds = mn * 12 + 16 + mn * (D // 4 + 1) * 16
+            if lds <= 160 * 1024:
+                aos, ncount, extra = _device_forest(
+                    model, X.device, v4_key=("eif_dense", D))
+                return ext.score_extended_dense_v2(
+                    X.contiguous(), aos, extra["values"], extra["hw"],
+                    ncount, extra["height"], c, finalize,
+                )
         # deep forests overflow the dense kernel's LDS weight staging:
         # fall through to the general kernel (nodes from global if needed)
     aos, ncount, extra = _device_forest(model, X.device)

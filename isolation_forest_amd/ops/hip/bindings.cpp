@@ -392,8 +392,8 @@ torch::Tensor score_extended_dense_v2(torch::Tensor X,
               "hw must be float32 [T, max_nodes, d]");
   int64_t N = X.size(0), d = X.size(1);
   int64_t T = nodes_packed.size(0), max_nodes = nodes_packed.size(1);
-  TORCH_CHECK(d <= 32, "dense v2 supports d <= 32");
-  const int D = d <= 8 ? 8 : (d <= 16 ? 16 : 32);
+  TORCH_CHECK(d <= 64, "dense v2 supports d <= 64");
+  const int D = d <= 8 ? 8 : (d <= 16 ? 16 : (d <= 32 ? 32 : 64));
   TORCH_CHECK(hw.size(2) == D, "hw must be host-padded to D columns");
   auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
   if (N == 0) return out;
@@ -434,8 +434,10 @@ torch::Tensor score_extended_dense_v3(torch::Tensor X,
               "hwp must be int32 [T, max_nodes, D/2] packed bf16 pairs");
   int64_t N = X.size(0), d = X.size(1);
   int64_t T = nodes_packed.size(0), max_nodes = nodes_packed.size(1);
-  TORCH_CHECK(d <= 32, "dense v3 supports d <= 32");
-  const int D = d <= 8 ? 8 : (d <= 16 ? 16 : 32);
+  TORCH_CHECK(d <= 128, "dense v3 supports d <= 128");
+  const int D = d <= 8 ? 8
+                       : (d <= 16 ? 16
+                                  : (d <= 32 ? 32 : (d <= 64 ? 64 : 128)));
   TORCH_CHECK(hwp.size(2) == D / 2, "hwp must be packed to D/2 dwords");
   auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
   if (N == 0) return out;
@@ -444,7 +446,7 @@ torch::Tensor score_extended_dense_v3(torch::Tensor X,
   TORCH_CHECK(lds <= kMaxLds, "tree too large for LDS staging");
   const char* rpt_env = getenv("IFA_EIF_V3_RPT2");
   const bool rpt2 = rpt_env && rpt_env[0] == '1';
-  const int rows_per_iter = (D == 32 && !rpt2) ? 512 : 1024;
+  const int rows_per_iter = (D >= 64 || (D == 32 && !rpt2)) ? 512 : 1024;
   int blocks = (int)std::min<int64_t>(
       (N + rows_per_iter - 1) / rows_per_iter, 8192);
   ifa::launch_score_extended_dense_v3(

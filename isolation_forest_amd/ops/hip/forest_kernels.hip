@@ -829,7 +829,8 @@ struct RowReg {
 #define EIFD_THREADS 512
 
 template <typename KT, int D, int RPT>
-__global__ void __launch_bounds__(EIFD_THREADS, 4) score_extended_dense_v2(
+__global__ void __launch_bounds__(EIFD_THREADS,
+                                  (D == 64) ? 2 : 4) score_extended_dense_v2(
     const KT* __restrict__ X,           // raw bits [N][d]
     const int2* __restrict__ nodes,     // [T][max_nodes] {w0, offset/-inf}
     const float* __restrict__ values,   // [T][max_nodes] leaf value+depth
@@ -893,7 +894,7 @@ __global__ void __launch_bounds__(EIFD_THREADS, 4) score_extended_dense_v2(
         // ended (leaf self-loop: right == own id) are exec-masked out of
         // the dot — masked lanes issue no LDS accesses, cutting the bank-
         // conflict cycles that bound this kernel.
-        constexpr int CHUNKS = (D >= 32) ? 4 : 2;
+        constexpr int CHUNKS = (D >= 64) ? 8 : ((D >= 32) ? 4 : 2);
         constexpr int CW4 = D / (4 * CHUNKS);  // float4s per chunk
 #pragma unroll
         for (int r = 0; r < RPT; ++r) {
@@ -972,7 +973,11 @@ __device__ __forceinline__ float dot2_bf16(uint32_t w, uint32_t x, float acc) {
 
 template <int D, int RPT>
 __global__ void __launch_bounds__(EIFD_THREADS,
-                                  (D == 32 && RPT == 2) ? 4 : 6)
+                                  (D == 128) ? 2
+                                             : ((D == 64
+                                                 || (D == 32 && RPT == 2))
+                                                    ? 4
+                                                    : 6))
 score_extended_dense_v3(
     const uint16_t* __restrict__ X,     // raw bf16 bits [N][d]
     const int2* __restrict__ nodes,     // [T][max_nodes] {right<<12, offset/-inf}
@@ -1548,10 +1553,12 @@ void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
   if (bf16) {
     if (D == 8) LSD2(uint16_t, 8, 2);
     else if (D == 16) LSD2(uint16_t, 16, 2);
+    else if (D == 64) LSD2(uint16_t, 64, 1);
     else LSD2(uint16_t, 32, 1);
   } else {
     if (D == 8) LSD2(uint32_t, 8, 2);
     else if (D == 16) LSD2(uint32_t, 16, 2);
+    else if (D == 64) LSD2(uint32_t, 64, 1);
     else LSD2(uint32_t, 32, 1);
   }
 #undef LSD2
@@ -1609,8 +1616,10 @@ void launch_score_extended_dense_v3(int D, bool rpt2, const void* X,
   } while (0)
   if (D == 8) LSD3(8, 2);
   else if (D == 16) LSD3(16, 2);
-  else if (rpt2) LSD3(32, 2);  // A/B: halves per-row staging amortization
-  else LSD3(32, 1);
+  else if (D == 32 && rpt2) LSD3(32, 2);  // A/B: halved staging amortization
+  else if (D == 32) LSD3(32, 1);
+  else if (D == 64) LSD3(64, 1);
+  else LSD3(128, 1);
 #undef LSD3
 }
 

@@ -464,14 +464,82 @@ class TestWideFeatureFallbacks:
             s_gpu.view(np.int32), s_cpu.view(np.int32))
 
     def test_extended_wide_d_wide_nnz(self, dev):
-        """d=40 > 32 with nnz=40 routes to the general strict-order
-        kernel (no dense v2): bitwise."""
+        """d=40 > 32 with nnz=40: round 2 routes this to the D=64 dense
+        band (f32 rows -> v2) — reassociation-tolerance contract."""
         X = torch.from_numpy(make_data(15000, 40, seed=43)).to(dev)
         model = ExtendedIsolationForest(numEstimators=12, randomSeed=6).fit(X)
         s_gpu = model.score(X).cpu().numpy()
         s_cpu = model.score(X.float().cpu()).numpy()
         diff = np.abs(s_gpu - s_cpu)
         assert np.quantile(diff, 0.999) < 1e-3
+
+
+class TestWideDenseBands:
+    """Round-2 dense bands D=64/128: fully-extended EIF beyond d=32 no
+    longer falls to the general kernel (which measured 8.2M rows/s at
+    d=64 — profiles/r02_eif_v3.md)."""
+
+    def _parity(self, dev, d, dtype):
+        import copy
+
+        from isolation_forest_amd.ops import gpu_engine
+        from isolation_forest_amd.utils.det_math import bf16_round
+
+        rs = np.random.RandomState(50 + d)
+        X32 = rs.normal(size=(3000, d)).astype(np.float32)
+        if dtype == torch.bfloat16:
+            Xt = torch.from_numpy(X32).to(torch.bfloat16)
+            X = Xt.float().numpy()
+        else:
+            Xt = torch.from_numpy(X32)
+            X = X32
+        bag = cpu_engine.sample_bags(3000, 8, 128, seed=d, bootstrap=False)
+        fs = cpu_engine.feature_subsets(d, d, 8, seed=d)
+        forest = cpu_engine.build_extended_forest(X, bag, fs, d, 128, d, d,
+                                                  d - 1)
+        if dtype == torch.bfloat16:
+            f_b = copy.copy(forest)
+            f_b.hyper_w = bf16_round(forest.hyper_w)
+            cpu_ps = cpu_engine.path_lengths_extended(f_b, X)
+        else:
+            cpu_ps = cpu_engine.path_lengths_extended(forest, X)
+        model = ExtendedIsolationForest(numEstimators=8).fit(X[:500])
+        model.forest = forest
+        model._gpu_forest_cache = {}
+        gpu_ps = gpu_engine.score_extended_forest(model, Xt.to(dev),
+                                                  finalize=False)
+        diff = np.abs(gpu_ps.cpu().numpy() - cpu_ps)
+        assert (diff > 1e-3).sum() <= 5
+        assert np.quantile(diff, 0.99) < 1e-3
+
+    def test_d64_bf16_v3(self, dev):
+        self._parity(dev, 64, torch.bfloat16)
+
+    def test_d64_f32_v2(self, dev):
+        self._parity(dev, 64, torch.float32)
+
+    def test_d128_bf16_v3(self, dev):
+        self._parity(dev, 128, torch.bfloat16)
+
+    def test_d128_f32_general_stays_bitwise(self, dev):
+        """f32 rows at d=128 exceed v2's LDS: the general strict-order
+        kernel keeps the bitwise contract."""
+        from isolation_forest_amd.ops import gpu_engine
+
+        rs = np.random.RandomState(99)
+        X = rs.normal(size=(2000, 128)).astype(np.float32)
+        bag = cpu_engine.sample_bags(2000, 4, 128, seed=9, bootstrap=False)
+        fs = cpu_engine.feature_subsets(128, 128, 4, seed=9)
+        forest = cpu_engine.build_extended_forest(X, bag, fs, 9, 128, 128,
+                                                  128, 127)
+        cpu_ps = cpu_engine.path_lengths_extended(forest, X)
+        model = ExtendedIsolationForest(numEstimators=4).fit(X[:500])
+        model.forest = forest
+        model._gpu_forest_cache = {}
+        gpu_ps = gpu_engine.score_extended_forest(
+            model, torch.from_numpy(X).to(dev), finalize=False)
+        np.testing.assert_array_equal(
+            gpu_ps.cpu().numpy().view(np.int32), cpu_ps.view(np.int32))
 
 
 class TestNativeCLI:

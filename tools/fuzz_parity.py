@@ -112,10 +112,21 @@ def one_case(rs: np.random.RandomState, it: int) -> str:
         sparse_route = (nnz <= 5
                         and mn * (12 + nnz * 8) + 2 * 256 * dpad_s * elem
                         <= 150 * 1024)
-        D = 8 if d <= 8 else (16 if d <= 16 else 32)
-        dense_route = (not sparse_route and d <= 32
+        if d <= 8:
+            D = 8
+        elif d <= 16:
+            D = 16
+        elif d <= 32:
+            D = 32
+        elif d <= 64:
+            D = 64
+        else:
+            D = 128
+        d_cap = 128 if bf16 else 64  # v3 packs bf16 weights; v2 f32
+        wbytes = (D // 8 + 1) * 16 if bf16 else (D // 4 + 1) * 16
+        dense_route = (not sparse_route and d <= d_cap
                        and (nnz == d or nnz >= 6)
-                       and mn * 12 + 16 + mn * (D // 4 + 1) * 16 <= 160 * 1024)
+                       and mn * 12 + 16 + mn * wbytes <= 160 * 1024)
         old_dense_route = (not sparse_route and not dense_route
                            and nnz == d and d % 4 == 0 and mn * 8 <= 120 * 1024)
         if dense_route or old_dense_route:
