@@ -264,24 +264,43 @@ torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
   }
   // deep forests (large maxSamples) overflow LDS node staging: walk nodes
   // from global/L2 instead (NODES_LDS=false) — correct for any tree size
-  size_t node_bytes = (size_t)ilp * max_nodes * 8;  // staged trees
-  bool nodes_lds = true;
-  if (node_bytes + (size_t)256 * dpad * elem > 150 * 1024) {
-    nodes_lds = false;
-    node_bytes = 0;
-  }
-  // prefer 2 rows/thread when 3 blocks/CU still fit, else 1, else global X
+  bool nodes_lds =
+      (size_t)2 * max_nodes * 8 + (size_t)256 * dpad * elem <= 150 * 1024;
+  // config search over (rpt, ilp): wide-d rows inflate the LDS row tile,
+  // so dropping staged-tree ILP to 2 can buy back a resident block
+  // (measured: d=128 went 1 block/CU at ILP=4 -> 124M rows/s; ILP=2 at 2
+  // blocks is the faster shape). Prefer the candidate with the most
+  // resident blocks, breaking ties toward more chains (rpt*ilp).
   int rpt = 2;
   bool rows_lds = true;
-  size_t lds;
-  for (;;) {
-    size_t row_bytes = (size_t)rpt * 256 * dpad * elem;
-    lds = node_bytes + row_bytes;
-    if (lds * 3 <= kMaxLds || (rpt == 1 && lds <= 150 * 1024)) break;
-    if (rpt == 2) { rpt = 1; continue; }
-    rows_lds = false;
-    lds = node_bytes;
-    break;
+  size_t lds = 0;
+  {
+    struct Cand { int rpt, ilp; };
+    const Cand cands[] = {{2, ilp}, {1, ilp}, {2, 2}, {1, 2}};
+    int best_blocks = 0, best_chains = 0;
+    bool found = false;
+    for (const Cand& cnd : cands) {
+      size_t nb = nodes_lds ? (size_t)cnd.ilp * max_nodes * 8 : 0;
+      size_t l = nb + (size_t)cnd.rpt * 256 * dpad * elem;
+      if (l > 150 * 1024) continue;
+      int blocks_cu = (int)std::min<size_t>(kMaxLds / std::max(l, (size_t)1),
+                                            3);
+      int chains = cnd.rpt * cnd.ilp;
+      if (blocks_cu > best_blocks ||
+          (blocks_cu == best_blocks && chains > best_chains)) {
+        best_blocks = blocks_cu;
+        best_chains = chains;
+        rpt = cnd.rpt;
+        ilp = cnd.ilp;
+        lds = l;
+        found = true;
+      }
+    }
+    if (!found) {  // rows exceed LDS even alone: global-X walk
+      rows_lds = false;
+      rpt = 1;
+      lds = nodes_lds ? (size_t)ilp * max_nodes * 8 : 0;
+    }
   }
   if (const char* e = getenv("IFA_SCORE_FORCE_GLOBAL")) {
     if (atoi(e)) { rows_lds = false; rpt = 1; lds = node_bytes; }

@@ -1488,7 +1488,9 @@ void launch_score_forest(bool bf16, int rpt, bool rows_lds, bool nodes_lds,
   } while (0)
 #define LS_ILP(KT, RPT, RL)                                                   \
   do {                                                                        \
-    if (ilp == 8) LS(KT, RPT, RL, 8); else LS(KT, RPT, RL, 4);                \
+    if (ilp == 8) LS(KT, RPT, RL, 8);                                         \
+    else if (ilp == 2) LS(KT, RPT, RL, 2);                                    \
+    else LS(KT, RPT, RL, 4);                                                  \
   } while (0)
   if (bf16) {
     if (rows_lds && rpt == 2) LS_ILP(uint16_t, 2, true);
