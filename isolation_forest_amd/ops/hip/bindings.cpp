@@ -303,7 +303,11 @@ torch::Tensor score_forest(torch::Tensor X, torch::Tensor nodes_packed,
     }
   }
   if (const char* e = getenv("IFA_SCORE_FORCE_GLOBAL")) {
-    if (atoi(e)) { rows_lds = false; rpt = 1; lds = node_bytes; }
+    if (atoi(e)) {
+      rows_lds = false;
+      rpt = 1;
+      lds = nodes_lds ? (size_t)ilp * max_nodes * 8 : 0;
+    }
   }
   int64_t rows_per_block = (int64_t)(rows_lds ? rpt : 1) * 256;
   int blocks = (int)std::min<int64_t>(
