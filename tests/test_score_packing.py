@@ -361,9 +361,8 @@ class TestEif0MirrorPack:
         from isolation_forest_amd.ops.gpu_engine import (
             _eif0_packed_v4, _node_depths)
 
-        forest, X = self._ext0_forest(7, foreign=True)
-        packed, ncount, max_depth = _eif0_packed_v4(forest, 5, False)
+        forest, _ = self._ext0_forest(7, foreign=True)
+        _, _, max_depth = _eif0_packed_v4(forest, 5, False)
         depth = _node_depths(forest.feature, forest.right)
-        live = forest.feature >= -1
         assert max_depth == int(depth[(forest.feature >= 0)
                                       | (forest.feature == -1)].max())
