@@ -149,23 +149,28 @@ def feature_subsets(
 
 def _split_feature(
     bag: np.ndarray, seg: np.ndarray, features: np.ndarray, seed: int,
-    tree_id: int, node_id: int,
+    tree_id: int, node_id: int, featsel_row: np.ndarray = None,
 ) -> Tuple[int, float, float]:
     """Draw a split feature with constant-feature retry.
 
     Fisher-Yates over the tree's feature subset; first feature whose values
     in this node are non-constant wins (IsolationTree.scala:124-150).
     Returns (feature, min, max) or (-1, 0, 0) when every feature is constant.
+    ``featsel_row`` optionally supplies the node's precomputed Fisher-Yates
+    offsets (featsel_table; bitwise-identical draws, vectorized Philox).
     """
     avail = features.copy()
     k = len(avail)
     for j in range(k):
-        t = j + int(
-            rng.randint_below(
-                seed, rng.P_FEATSEL, np.uint32(tree_id), np.uint32(node_id),
-                k - j, attempt=np.uint32(j),
+        if featsel_row is not None:
+            t = j + int(featsel_row[j])
+        else:
+            t = j + int(
+                rng.randint_below(
+                    seed, rng.P_FEATSEL, np.uint32(tree_id),
+                    np.uint32(node_id), k - j, attempt=np.uint32(j),
+                )
             )
-        )
         avail[j], avail[t] = avail[t], avail[j]
         f = int(avail[j])
         col = bag[seg, f]
@@ -174,6 +179,25 @@ def _split_feature(
         if fmin < fmax:
             return f, float(fmin), float(fmax)
     return -1, 0.0, 0.0
+
+
+def draw_tables(seed: int, tree_ids: np.ndarray, max_nodes: int, k: int):
+    """Vectorized per-(tree, node) Philox draws for the standard build.
+
+    Bitwise-identical to the per-node scalar calls build_tree would
+    otherwise make (same keys/purposes; rng.py is broadcast-capable) —
+    one batched Philox evaluation replaces ~2*nodes scalar ones, the
+    dominant cost of the CPU oracle build (profiled 80%).
+    Returns (usplit [T, mn] float64, featsel [T, mn, k] int64)."""
+    T = len(tree_ids)
+    tg = tree_ids.astype(np.uint32)[:, None]
+    ng = np.arange(max_nodes, dtype=np.uint32)[None, :]
+    usplit = rng.uniform(seed, rng.P_SPLIT, tg, ng)
+    featsel = np.empty((T, max_nodes, k), dtype=np.int64)
+    for j in range(k):
+        featsel[:, :, j] = rng.randint_below(
+            seed, rng.P_FEATSEL, tg, ng, k - j, attempt=np.uint32(j))
+    return usplit, featsel
 
 
 def _split_value32(fmin: float, fmax: float, u: float) -> np.float32:
@@ -195,6 +219,9 @@ def build_tree(
     out_right: np.ndarray,
     out_count: np.ndarray,
     out_value64: np.ndarray = None,
+    usplit_row: np.ndarray = None,   # [max_nodes] precomputed split uniforms
+    featsel_tab: np.ndarray = None,  # [max_nodes, k] precomputed FY offsets
+    leaf_lut: np.ndarray = None,     # [n+1] precomputed c(m) float32
 ) -> int:
     """Build one iTree into pre-order SoA rows; returns node count."""
     n = bag.shape[0]
@@ -214,16 +241,22 @@ def build_tree(
         m = len(seg)
         if m <= 1 or height >= height_limit:
             out_feature[node] = Forest.LEAF
-            out_value[node] = avg_path_length(m)
+            out_value[node] = (leaf_lut[m] if leaf_lut is not None
+                               else avg_path_length(m))
             out_count[node] = m
             continue
-        f, fmin, fmax = _split_feature(bag, seg, features, seed, tree_id, node)
+        f, fmin, fmax = _split_feature(
+            bag, seg, features, seed, tree_id, node,
+            featsel_tab[node] if featsel_tab is not None else None)
         if f < 0:
             out_feature[node] = Forest.LEAF
-            out_value[node] = avg_path_length(m)
+            out_value[node] = (leaf_lut[m] if leaf_lut is not None
+                               else avg_path_length(m))
             out_count[node] = m
             continue
-        u = float(rng.uniform(seed, rng.P_SPLIT, np.uint32(tree_id), np.uint32(node)))
+        u = (float(usplit_row[node]) if usplit_row is not None
+             else float(rng.uniform(seed, rng.P_SPLIT, np.uint32(tree_id),
+                                    np.uint32(node))))
         s32 = _split_value32(fmin, fmax, u)
         mask = bag[seg, f] < s32
         left = seg[mask]
@@ -252,6 +285,10 @@ def build_forest(
     T, n = bag_idx.shape
     max_nodes = 2 * n - 1 if n >= 1 else 1
     forest = empty_forest(T, max_nodes, num_samples, num_features, total_num_features)
+    tree_ids = np.arange(T, dtype=np.int64) + tree_id_offset
+    k = feat_sub.shape[1]
+    usplit, featsel = draw_tables(seed, tree_ids, max_nodes, k)
+    leaf_lut = avg_path_length(np.arange(n + 1)).astype(np.float32)
     for t in range(T):
         bag = X[bag_idx[t]]
         nc = build_tree(
@@ -264,6 +301,9 @@ def build_forest(
             forest.right[t],
             forest.num_instances[t],
             forest.value64[t],
+            usplit_row=usplit[t],
+            featsel_tab=featsel[t],
+            leaf_lut=leaf_lut,
         )
         forest.node_count[t] = nc
     return forest
