@@ -321,34 +321,60 @@ def _gaussian(seed: int, tree_id: int, counter: int) -> float:
     return float(det_math.det_gaussian(u1, u2))
 
 
+def draw_tables_extended(seed: int, tree_ids: np.ndarray, max_nodes: int,
+                         k: int, nnz: int):
+    """Vectorized per-(tree, node, slot) Philox draws for the EIF build —
+    bitwise-identical counters to the per-node calls in _draw_hyperplane
+    (same purposes/keys, rng.py broadcasts). Returns
+    (fy [T,mn,nnz] int64, w32 [T,mn,nnz] float32 unnormalized gaussians,
+    uint [T,mn,nnz] float64 intercept uniforms)."""
+    tg = tree_ids.astype(np.uint32)[:, None, None]
+    ng = np.arange(max_nodes, dtype=np.int64)[None, :, None]
+    jarr = np.arange(nnz, dtype=np.int64)[None, None, :]
+    fy = rng.randint_below(
+        seed, rng.P_EIF_COORD, tg, ng.astype(np.uint32),
+        (k - jarr), attempt=jarr.astype(np.uint32),
+    )
+    slot = (ng * 4096 + jarr).astype(np.uint32)
+    u1, u2 = rng.uniform2(seed, rng.P_EIF_NORMAL, tg, slot)
+    w32 = det_math.det_gaussian(u1, u2).astype(np.float32)
+    uint = rng.uniform(seed, rng.P_EIF_INTERCEPT, tg, slot).astype(np.float64)
+    return fy, w32, uint
+
+
 def _draw_hyperplane(
     bag: np.ndarray, seg: np.ndarray, features: np.ndarray, nnz: int,
-    seed: int, tree_id: int, node_id: int,
+    seed: int, tree_id: int, node_id: int, tabs=None,
 ):
     """Returns (idx[nnz] int32 ascending, w[nnz] float32, offset64, offset32)
     or None when the weight vector has zero norm (theoretical leaf case,
     ExtendedIsolationTree.scala:183-184)."""
     k = len(features)
     avail = features.copy()
-    jarr = np.arange(nnz, dtype=np.uint32)
-    # batched Philox draws — identical counters (and therefore bitwise-
-    # identical values) to per-j scalar calls, ~5x faster per node
-    fy = rng.randint_below(
-        seed, rng.P_EIF_COORD, np.uint32(tree_id), np.uint32(node_id),
-        (k - jarr.astype(np.int64)), attempt=jarr,
-    )
+    if tabs is not None:
+        fy = tabs[0][node_id]
+        w = tabs[1][node_id].copy()
+    else:
+        jarr = np.arange(nnz, dtype=np.uint32)
+        # batched Philox draws — identical counters (and therefore bitwise-
+        # identical values) to per-j scalar calls, ~5x faster per node
+        fy = rng.randint_below(
+            seed, rng.P_EIF_COORD, np.uint32(tree_id), np.uint32(node_id),
+            (k - jarr.astype(np.int64)), attempt=jarr,
+        )
     for j in range(nnz):
         t = j + int(fy[j])
         avail[j], avail[t] = avail[t], avail[j]
     coords = np.array(sorted(int(c) for c in avail[:nnz]), dtype=np.int32)
 
-    # weights drawn AFTER sorting, keyed by slot j over sorted coords
-    base = node_id * 4096  # nnz-slot stride; nnz <= 4096 by construction
-    u1, u2 = rng.uniform2(
-        seed, rng.P_EIF_NORMAL, np.uint32(tree_id),
-        np.uint32(base) + jarr,
-    )
-    w = det_math.det_gaussian(u1, u2).astype(np.float32)
+    if tabs is None:
+        # weights drawn AFTER sorting, keyed by slot j over sorted coords
+        base = node_id * 4096  # nnz-slot stride; nnz <= 4096 by construction
+        u1, u2 = rng.uniform2(
+            seed, rng.P_EIF_NORMAL, np.uint32(tree_id),
+            np.uint32(base) + jarr,
+        )
+        w = det_math.det_gaussian(u1, u2).astype(np.float32)
     # L2 normalize in float32, sequential accumulation
     acc = np.float32(0.0)
     for j in range(nnz):
@@ -362,10 +388,13 @@ def _draw_hyperplane(
     sub = bag[np.ix_(seg, coords)]
     mn = sub.min(axis=0).astype(np.float32).astype(np.float64)
     mx = sub.max(axis=0).astype(np.float32).astype(np.float64)
-    u = rng.uniform(
-        seed, rng.P_EIF_INTERCEPT, np.uint32(tree_id),
-        np.uint32(base) + jarr,
-    ).astype(np.float64)
+    if tabs is not None:
+        u = tabs[2][node_id]
+    else:
+        u = rng.uniform(
+            seed, rng.P_EIF_INTERCEPT, np.uint32(tree_id),
+            np.uint32(base) + jarr,
+        ).astype(np.float64)
     intercepts = mn + u * (mx - mn)
     off64 = 0.0
     for j in range(nnz):
@@ -390,6 +419,8 @@ def build_extended_tree(
     tree_id: int,
     fr: ExtendedForest,
     t: int,
+    tabs=None,
+    leaf_lut: np.ndarray = None,
 ) -> int:
     n = bag.shape[0]
     height_limit = int(math.ceil(math.log2(max(n, 2))))
@@ -405,13 +436,16 @@ def build_extended_tree(
         m = len(seg)
         if m <= 1 or height >= height_limit:
             fr.feature[t, node] = ExtendedForest.LEAF
-            fr.value[t, node] = avg_path_length(m)
+            fr.value[t, node] = (leaf_lut[m] if leaf_lut is not None
+                                 else avg_path_length(m))
             fr.num_instances[t, node] = m
             continue
-        hp = _draw_hyperplane(bag, seg, features, nnz, seed, tree_id, node)
+        hp = _draw_hyperplane(bag, seg, features, nnz, seed, tree_id, node,
+                              tabs=tabs)
         if hp is None:
             fr.feature[t, node] = ExtendedForest.LEAF
-            fr.value[t, node] = avg_path_length(m)
+            fr.value[t, node] = (leaf_lut[m] if leaf_lut is not None
+                                 else avg_path_length(m))
             fr.num_instances[t, node] = m
             continue
         coords, w, off64, off32 = hp
@@ -450,10 +484,15 @@ def build_extended_forest(
         T, max_nodes, nnz, num_samples, num_features, total_num_features,
         extension_level,
     )
+    tree_ids = np.arange(T, dtype=np.int64) + tree_id_offset
+    k = feat_sub.shape[1]
+    fy_t, w_t, u_t = draw_tables_extended(seed, tree_ids, max_nodes, k, nnz)
+    leaf_lut = avg_path_length(np.arange(n + 1)).astype(np.float32)
     for t in range(T):
         bag = X[bag_idx[t]]
         fr.node_count[t] = build_extended_tree(
-            bag, feat_sub[t], nnz, seed, t + tree_id_offset, fr, t
+            bag, feat_sub[t], nnz, seed, t + tree_id_offset, fr, t,
+            tabs=(fy_t[t], w_t[t], u_t[t]), leaf_lut=leaf_lut,
         )
     return fr
 
