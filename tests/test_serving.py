@@ -6,6 +6,7 @@ validation, extended-model auto-detection, and serving a reference
 Spark-written golden model.
 """
 
+import json
 import numpy as np
 import pytest
 import torch
@@ -92,3 +93,51 @@ class TestServing:
         scores = r.json()["scores"]
         assert len(scores) == 16
         assert all(0.0 < s < 1.0 for s in scores)
+
+
+class TestServingProcess:
+    def test_uvicorn_process_end_to_end(self, saved_model):
+        """Real server process (python -m isolation_forest_amd.serving),
+        real TCP socket — the deployment path, not just the ASGI shim."""
+        import os
+        import socket
+        import subprocess
+        import sys
+        import time
+        import urllib.request
+
+        _, X, path = saved_model
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+        s.close()
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "isolation_forest_amd.serving",
+             "--model", path, "--device", "cpu", "--port", str(port)],
+            cwd=repo, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        )
+        try:
+            deadline = time.time() + 60
+            while True:
+                try:
+                    with urllib.request.urlopen(
+                            f"http://127.0.0.1:{port}/healthz", timeout=2) as r:
+                        assert r.status == 200
+                        break
+                except Exception:
+                    if time.time() > deadline or proc.poll() is not None:
+                        out = proc.stdout.read().decode(errors="replace")
+                        raise AssertionError(f"server failed to start: {out[-2000:]}")
+                    time.sleep(0.25)
+            body = json.dumps({"instances": X[:4].tolist()}).encode()
+            req = urllib.request.Request(
+                f"http://127.0.0.1:{port}/v1/score", data=body,
+                headers={"Content-Type": "application/json"})
+            with urllib.request.urlopen(req, timeout=10) as r:
+                resp = json.loads(r.read())
+            assert len(resp["scores"]) == 4
+            assert all(0.0 < s < 1.0 for s in resp["scores"])
+        finally:
+            proc.terminate()
+            proc.wait(timeout=20)
