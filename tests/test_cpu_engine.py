@@ -282,3 +282,43 @@ class TestExtendedBuild:
         s = cpu_engine.score_extended_forest(f, X)
         assert s.min() > 0.0 and s.max() < 1.0
         assert s[y == 1].mean() > s[y == 0].mean() + 0.1
+
+
+class TestFastWalkEquivalence:
+    """_path_lengths_fast (vectorized fixed-trip walk) must be bitwise-
+    identical to the masked reference walk, including foreign f64
+    knife-edge splits and NaN rows."""
+
+    def test_bitwise_on_random_and_adversarial(self):
+        rs = np.random.RandomState(17)
+        for trial, (N, d, n, T) in enumerate(
+                [(3000, 6, 256, 30), (1200, 3, 64, 10), (800, 12, 128, 7)]):
+            X = rs.normal(size=(N, d)).astype(np.float32)
+            bag = cpu_engine.sample_bags(N, T, n, seed=trial, bootstrap=False)
+            fs = cpu_engine.feature_subsets(d, d, T, seed=trial)
+            f = cpu_engine.build_forest(X, bag, fs, trial, n, d, d)
+            if trial == 1:  # foreign knife-edge splits
+                internal = f.feature >= 0
+                v32 = f.value.astype(np.float64)
+                nxt = np.nextafter(f.value,
+                                   np.float32(np.inf)).astype(np.float64)
+                f.value64 = np.where(internal, (v32 + nxt) / 2, f.value64)
+                X[:200, 0] = f.value[0, 0]
+            if trial == 2:
+                X[rs.randint(0, N, 40), rs.randint(0, d, 40)] = np.nan
+            a = cpu_engine._path_lengths_fast(f, X)
+            b = cpu_engine._path_lengths_ref(f, X)
+            np.testing.assert_array_equal(
+                a.view(np.int32), b.view(np.int32), err_msg=f"trial {trial}")
+
+    def test_env_forces_reference(self, monkeypatch):
+        rs = np.random.RandomState(18)
+        X = rs.normal(size=(500, 4)).astype(np.float32)
+        bag = cpu_engine.sample_bags(500, 4, 64, seed=1, bootstrap=False)
+        fs = cpu_engine.feature_subsets(4, 4, 4, seed=1)
+        f = cpu_engine.build_forest(X, bag, fs, 1, 64, 4, 4)
+        monkeypatch.setenv("IFA_CPU_WALK_REF", "1")
+        a = cpu_engine.path_lengths(f, X)
+        monkeypatch.delenv("IFA_CPU_WALK_REF")
+        b = cpu_engine.path_lengths(f, X)
+        np.testing.assert_array_equal(a.view(np.int32), b.view(np.int32))
