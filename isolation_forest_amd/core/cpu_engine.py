@@ -285,27 +285,32 @@ def build_forest(
     T, n = bag_idx.shape
     max_nodes = 2 * n - 1 if n >= 1 else 1
     forest = empty_forest(T, max_nodes, num_samples, num_features, total_num_features)
-    tree_ids = np.arange(T, dtype=np.int64) + tree_id_offset
     k = feat_sub.shape[1]
-    usplit, featsel = draw_tables(seed, tree_ids, max_nodes, k)
     leaf_lut = avg_path_length(np.arange(n + 1)).astype(np.float32)
-    for t in range(T):
-        bag = X[bag_idx[t]]
-        nc = build_tree(
-            bag,
-            feat_sub[t],
-            seed,
-            t + tree_id_offset,
-            forest.feature[t],
-            forest.value[t],
-            forest.right[t],
-            forest.num_instances[t],
-            forest.value64[t],
-            usplit_row=usplit[t],
-            featsel_tab=featsel[t],
-            leaf_lut=leaf_lut,
-        )
-        forest.node_count[t] = nc
+    # chunk the draw tables so memory stays bounded at large tree counts
+    # (featsel is [chunk, max_nodes, k] int64)
+    chunk = max(1, min(T, int(64e6 // max(max_nodes * k, 1)) or 1))
+    for t0 in range(0, T, chunk):
+        t1 = min(T, t0 + chunk)
+        tree_ids = np.arange(t0, t1, dtype=np.int64) + tree_id_offset
+        usplit, featsel = draw_tables(seed, tree_ids, max_nodes, k)
+        for t in range(t0, t1):
+            bag = X[bag_idx[t]]
+            nc = build_tree(
+                bag,
+                feat_sub[t],
+                seed,
+                t + tree_id_offset,
+                forest.feature[t],
+                forest.value[t],
+                forest.right[t],
+                forest.num_instances[t],
+                forest.value64[t],
+                usplit_row=usplit[t - t0],
+                featsel_tab=featsel[t - t0],
+                leaf_lut=leaf_lut,
+            )
+            forest.node_count[t] = nc
     return forest
 
 
@@ -484,16 +489,21 @@ def build_extended_forest(
         T, max_nodes, nnz, num_samples, num_features, total_num_features,
         extension_level,
     )
-    tree_ids = np.arange(T, dtype=np.int64) + tree_id_offset
     k = feat_sub.shape[1]
-    fy_t, w_t, u_t = draw_tables_extended(seed, tree_ids, max_nodes, k, nnz)
     leaf_lut = avg_path_length(np.arange(n + 1)).astype(np.float32)
-    for t in range(T):
-        bag = X[bag_idx[t]]
-        fr.node_count[t] = build_extended_tree(
-            bag, feat_sub[t], nnz, seed, t + tree_id_offset, fr, t,
-            tabs=(fy_t[t], w_t[t], u_t[t]), leaf_lut=leaf_lut,
-        )
+    chunk = max(1, min(T, int(32e6 // max(max_nodes * max(nnz, 1), 1)) or 1))
+    for t0 in range(0, T, chunk):
+        t1 = min(T, t0 + chunk)
+        tree_ids = np.arange(t0, t1, dtype=np.int64) + tree_id_offset
+        fy_t, w_t, u_t = draw_tables_extended(seed, tree_ids, max_nodes, k,
+                                              nnz)
+        for t in range(t0, t1):
+            bag = X[bag_idx[t]]
+            fr.node_count[t] = build_extended_tree(
+                bag, feat_sub[t], nnz, seed, t + tree_id_offset, fr, t,
+                tabs=(fy_t[t - t0], w_t[t - t0], u_t[t - t0]),
+                leaf_lut=leaf_lut,
+            )
     return fr
 
 
