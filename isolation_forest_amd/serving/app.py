@@ -40,8 +40,10 @@ def create_app(model_path: str, device: str = None) -> FastAPI:
         device = "cuda:0" if torch.cuda.is_available() else "cpu"
     dev = torch.device(device)
     model = load_any_model(model_path)
-    d = model.total_num_features if model.total_num_features > 0 \
-        else model.num_features
+    # dimension validation only when totalNumFeatures is known — legacy
+    # models carry the -1 sentinel and accept any width, exactly like the
+    # reference's transform (IsolationForestModel.scala:132-134)
+    d = model.total_num_features if model.total_num_features > 0 else -1
     threshold = model.outlier_score_threshold
 
     app = FastAPI(title="isolation-forest-amd", version="1.0")
@@ -79,8 +81,15 @@ def create_app(model_path: str, device: str = None) -> FastAPI:
                 status_code=400,
                 detail=f"each instance must have {d} features, "
                        f"got shape {list(X.shape)}")
-        with torch.no_grad():
-            scores = model.score(torch.from_numpy(X).to(dev))
+        try:
+            with torch.no_grad():
+                scores = model.score(torch.from_numpy(X).to(dev))
+        except (IndexError, RuntimeError, ValueError) as e:
+            # legacy models (no totalNumFeatures) skip up-front width
+            # validation, matching the reference; a width the trees cannot
+            # index still fails cleanly here
+            raise HTTPException(status_code=400,
+                                detail=f"scoring failed: {e}")
         scores_np = scores.float().cpu().numpy()
         labels = None
         if threshold >= 0:
