@@ -25,7 +25,8 @@ from isolation_forest_amd.utils.params import ResolvedParams
 
 def one_case(rs: np.random.RandomState, it: int) -> str:
     rows = int(rs.randint(300, 20000))
-    d = int(rs.choice([1, 2, 3, 5, 8, 13, 16, 31, 32, 47, 64]))
+    d = int(rs.choice([1, 2, 3, 5, 8, 13, 16, 31, 32, 47, 64, 100, 128,
+                       200]))
     n = int(rs.choice([2, 4, 16, 64, 128, 256, 512, 1024, 4096]))
     n = min(n, rows)
     k = int(rs.randint(1, d + 1))
