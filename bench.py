@@ -109,6 +109,10 @@ def main():
     ap.add_argument("--dtype", choices=["bf16", "fp32"], default="bf16")
     ap.add_argument("--contamination", type=float, default=0.0)
     ap.add_argument("--device", default=None)
+    ap.add_argument("--export-onnx", default=None, metavar="PATH",
+                    help="after timing, rank 0 exports the trained model's "
+                         "ONNX graph here (config #5's export leg; "
+                         "standard IF only)")
     args = ap.parse_args()
 
     if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
@@ -189,6 +193,12 @@ def main():
     # quality (outside the timed region): AUROC on this rank's shard
     sample = min(rows, 10_000_000)
     measured_auroc = auroc_torch(y[:sample], scores[:sample])
+
+    if rank == 0 and args.export_onnx and not args.extended:
+        from isolation_forest_amd.onnx import IsolationForestConverter
+
+        IsolationForestConverter.from_model(model).convert_and_save(
+            args.export_onnx)
 
     if rank == 0:
         result = {
