@@ -18,10 +18,13 @@ from isolation_forest_amd.onnx import IsolationForestConverter, evaluator
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--device", default="cuda:0" if torch.cuda.is_available() else "cpu")
-    ap.add_argument("--rows", type=int, default=1_000_000)
+    ap.add_argument("--rows", type=int, default=None,
+                    help="default: 1M on GPU, 100k on CPU")
     ap.add_argument("--features", type=int, default=32)
     ap.add_argument("--out", default="/tmp/ifa_example")
     args = ap.parse_args()
+    if args.rows is None:
+        args.rows = 1_000_000 if args.device.startswith("cuda") else 100_000
 
     g = torch.Generator(device=args.device).manual_seed(0)
     X = torch.randn((args.rows, args.features), device=args.device, generator=g)
