@@ -17,15 +17,26 @@
 //    2 rows x 4 staged trees = 8 chains/thread with batch-phased LDS reads
 //    (staggered lgkmcnt waits). Bitwise == cpu_engine.path_lengths.
 //  * score_extended_dense_v2      — K7 fast path (hyperplanes densified to
-//    D in {8,16,32} columns): rows resident in f32 REGISTERS across the
-//    whole tree loop, one tree's nodes+values+weights staged in LDS
-//    (conflict-breaking D/4+1 float4 row stride), exec-masked finished
+//    D in {8,16,32,64} columns; f32 rows): rows resident in f32 REGISTERS
+//    across the whole tree loop, one tree's nodes+values+weights staged in
+//    LDS (conflict-breaking D/4+1 float4 row stride), exec-masked finished
 //    lanes, offset=-inf self-looping leaves. Tolerance contract (4-partial
 //    fma dot), routing in ops/gpu_engine.score_extended_forest.
+//  * score_extended_dense_v3      — K7 fast path for bf16 rows, D up to
+//    128: weights staged as RNE-rounded PACKED bf16 (half the LDS bytes
+//    of v2 — its measured bound), rows as packed-bf16 register pairs,
+//    dot on v_dot2c_f32_bf16. Contract: PARITY.md "Numeric contracts".
 //  * score_extended_sparse_v2     — K7 small-nnz path (templated NNZ<=5):
 //    same fixed-trip structure, strict oracle j-order dot (bitwise).
+//    extensionLevel-0 forests do NOT use it: they are packed onto
+//    score_forest_v4 via exact key thresholds + mirrored subtrees
+//    (ops/gpu_engine._eif0_packed_v4 — bitwise, zero per-visit overhead).
 //  * score_extended_forest_kernel — K7 general strict-order fallback
 //    (wide d / ragged hyperplane widths).
+//  * score_forest_wide /
+//    score_extended_wide          — full-width int4 node records for
+//    forests past the packed 12-bit-feature/15-bit-node caps (foreign or
+//    CPU-built deep models); global-memory walks, correctness path.
 //  * bag_gather_kernel            — K9/K10 device side: gather sampled rows
 //    into per-tree bags (indices drawn host-side by the same Philox).
 //
