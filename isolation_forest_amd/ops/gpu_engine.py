@@ -11,8 +11,18 @@ are built ON DEVICE with torch ops from the raw build outputs (depths come
 from the build kernels) and cached per model+device; the numpy host pack
 is the fallback for loaded models and is bitwise-identical
 (tests/test_gpu.py::TestDevicePacking). EIF scoring routes by hyperplane
-width: nnz<=5 uniform -> sparse v2 (bitwise), else d<=32 -> densified
-dense v2 (tolerance), else the general strict-order kernel.
+width and dtype (score_extended_forest):
+
+* nnz == 1 (extensionLevel 0) -> the STANDARD v4 walk via exact key
+  thresholds + mirrored subtrees (_eif0_packed_v4; bitwise; NaN rows
+  fall through to the strict kernels),
+* nnz <= 5 uniform -> sparse v2 (bitwise strict j-order),
+* densifiable (nnz == d or nnz >= 6): bf16 rows -> dense v3 (packed
+  bf16 weights, d <= 128), f32 rows -> dense v2 (d <= 64) — tolerance
+  contracts per PARITY.md,
+* otherwise (wide d, ragged widths, LDS overflow) -> the general
+  strict-order kernel; forests past the packed 15-bit-node/12-bit-
+  feature caps use the wide int4 kernels.
 """
 
 from __future__ import annotations
