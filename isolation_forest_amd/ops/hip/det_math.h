@@ -82,7 +82,12 @@ __device__ __forceinline__ double det_gaussian(double u1, double u2) {
   if (u1 < 1.0) {
     double t = 1.0 - u1;
     if (t < 1e-300) t = 1e-300;
-    r = sqrt(-2.0 * det_log(t));
+    // det_log(1.0) is ~1e-12, not exactly 0 (polynomial residual), so the
+    // u1 == 0 draw would take sqrt of a negative and poison the node's
+    // weights with NaN (fuzz-found at 2^-24 probability per draw). The
+    // true Box-Muller radial term at u1 = 0 is 0: clamp the argument.
+    double a = -2.0 * det_log(t);
+    r = sqrt(a > 0.0 ? a : 0.0);
   }
   return r * det_cos2pi(u2);
 }

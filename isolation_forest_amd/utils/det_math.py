@@ -90,7 +90,14 @@ def bf16_round(a: np.ndarray) -> np.ndarray:
 
 
 def det_gaussian(u1, u2):
-    """Box-Muller with deterministic log/cos; returns float64."""
+    """Box-Muller with deterministic log/cos; returns float64.
+
+    det_log(1.0) carries a ~1e-12 polynomial residual rather than exact 0,
+    so the u1 == 0 draw (probability 2^-24) would take sqrt of a negative
+    and poison a hyperplane with NaN weights (fuzz-found). The true radial
+    term at u1 = 0 is 0: clamp the sqrt argument at 0 (mirrored in
+    ops/hip/det_math.h — bitwise CPU == GPU)."""
     u1 = np.asarray(u1, dtype=np.float64)
-    r = np.where(u1 < 1.0, np.sqrt(-2.0 * det_log(np.maximum(1.0 - u1, 1e-300))), 0.0)
+    a = -2.0 * det_log(np.maximum(1.0 - u1, 1e-300))
+    r = np.where(u1 < 1.0, np.sqrt(np.maximum(a, 0.0)), 0.0)
     return r * det_cos2pi(u2)

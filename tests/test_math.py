@@ -109,3 +109,27 @@ class TestBf16Round:
         assert bf16_round(np.array([tie]))[0] == np.float32(1.0)
         out = bf16_round(np.zeros((3, 4), dtype=np.float32))
         assert out.shape == (3, 4)
+
+
+class TestDetGaussianEdges:
+    def test_u1_zero_is_finite_zero(self):
+        """u1 == 0 (a 2^-24 Philox draw, fuzz-found) must give the true
+        Box-Muller radial term 0, not sqrt(negative) = NaN."""
+        from isolation_forest_amd.utils.det_math import det_gaussian
+
+        u1 = np.array([0.0, 2.0 ** -24, 0.5, 1.0 - 2.0 ** -24, 1.0 - 1e-16])
+        u2 = np.array([0.9186, 0.1, 0.2, 0.3, 0.77])
+        g = det_gaussian(u1, u2)
+        assert np.isfinite(g).all()
+        assert g[0] == 0.0
+
+    def test_grid_finite(self):
+        from isolation_forest_amd.utils.det_math import det_gaussian
+
+        rs = np.random.RandomState(0)
+        u1 = np.concatenate([rs.rand(20000),
+                             np.arange(16) * 2.0 ** -24,
+                             1.0 - np.arange(1, 16) * 2.0 ** -24])
+        u2 = rs.rand(len(u1))
+        g = det_gaussian(u1, u2)
+        assert np.isfinite(g).all()
