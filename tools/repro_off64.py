@@ -32,6 +32,36 @@ def main():
     rp = ResolvedParams(num_samples=n, num_features=k, total_rows=rows,
                         total_features=d, extension_level=ext_level)
     gpu_f = gpu_engine.build_extended_forest(Xt, bag, fs, seed, rp)
+    # localize any structural divergence
+    for name in ["feature", "right", "num_instances", "node_count"]:
+        a, b = getattr(cpu_f, name), getattr(gpu_f, name)
+        if not np.array_equal(a, b):
+            if a.ndim == 1:
+                print(f"{name} differs: cpu {a} gpu {b}")
+                continue
+            trees = sorted(set(np.argwhere(a != b)[:, 0].tolist()))
+            print(f"{name} differs in trees {trees[:10]}")
+            t0 = trees[0]
+            nodes = np.argwhere(a[t0] != b[t0]).ravel()
+            nd = int(nodes[0])
+            print(f"  tree {t0} first node {nd}: cpu={a[t0, nd]} "
+                  f"gpu={b[t0, nd]}")
+            nnz_ = cpu_f.nnz
+            # dump both sides around the first divergent node
+            for x in range(max(0, nd - 2), min(a.shape[1], nd + 3)):
+                print(f"   node {x}: cpu(feat={cpu_f.feature[t0,x]},"
+                      f" right={cpu_f.right[t0,x]}, cnt={cpu_f.num_instances[t0,x]},"
+                      f" off32={cpu_f.value[t0,x]!r})"
+                      f" gpu(feat={gpu_f.feature[t0,x]},"
+                      f" right={gpu_f.right[t0,x]}, cnt={gpu_f.num_instances[t0,x]},"
+                      f" off32={gpu_f.value[t0,x]!r})")
+                wc = cpu_f.hyper_w[t0, x, :nnz_]
+                wg = gpu_f.hyper_w[t0, x, :nnz_]
+                if not np.array_equal(wc.view(np.int32), wg.view(np.int32)):
+                    dj = np.nonzero(wc.view(np.int32) != wg.view(np.int32))[0]
+                    print(f"     hyper_w differs at slots {dj[:6]}: "
+                          f"cpu {wc[dj[:3]]} gpu {wg[dj[:3]]}")
+            return
     for name in ["node_count", "feature", "right", "hyper_idx"]:
         assert np.array_equal(getattr(gpu_f, name), getattr(cpu_f, name)), name
     assert np.array_equal(gpu_f.hyper_w.view(np.int32),
