@@ -52,7 +52,7 @@ def _knife_edge_explained(forest, x_row: np.ndarray) -> bool:
 def one_case(rs: np.random.RandomState, it: int) -> str:
     rows = int(rs.randint(300, 20000))
     d = int(rs.choice([1, 2, 3, 5, 8, 13, 16, 31, 32, 47, 64, 100, 128,
-                       200]))
+                       200, 600]))
     n = int(rs.choice([2, 4, 16, 64, 128, 256, 512, 1024, 4096]))
     n = min(n, rows)
     k = int(rs.randint(1, d + 1))
