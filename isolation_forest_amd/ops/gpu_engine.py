@@ -919,7 +919,7 @@ def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torc
             while dpad % 2 != 1:
                 dpad += 1
         mn = forest.feature.shape[1]
-        lds = mn * (12 + nnz * 8) + 8 + 2 * 256 * dpad * elem
+        lds = mn * (12 + nnz * 8) + 2 * 256 * dpad * elem
         if lds <= 150 * 1024:
             aos, ncount, extra = _device_forest(
                 model, X.device, v4_key="eif_sparse")

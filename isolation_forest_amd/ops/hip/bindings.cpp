@@ -516,8 +516,8 @@ torch::Tensor score_extended_sparse_v2(torch::Tensor X,
   } else {
     while (dpad % 2 != 1) ++dpad;
   }
-  size_t lds = (size_t)max_nodes * (12 + nnz * 8) + 8  // int2 align slack
-               + (size_t)2 * 256 * dpad * elem;
+  size_t lds = (size_t)max_nodes * (12 + nnz * 8)
+               + (size_t)2 * 256 * dpad * elem;  // iw int2 == old idx+w bytes
   TORCH_CHECK(lds <= 150 * 1024, "sparse v2 LDS overflow; use general path");
   int blocks = (int)std::min<int64_t>((N + 511) / 512, 8192);
   ifa::launch_score_extended_sparse_v2(

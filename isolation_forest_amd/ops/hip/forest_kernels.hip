@@ -1129,10 +1129,8 @@ __global__ void __launch_bounds__(256) score_extended_sparse_v2(
   int2* tlds = (int2*)smem;                   // [max_nodes]
   float* vlds = (float*)(tlds + max_nodes);   // [max_nodes]
   // (idx, weight) interleaved as int2: one ds_read_b64 per coordinate
-  // instead of two b32 reads (the visit is LDS-instruction-bound).
-  // MUST be 8-byte aligned: the base lands at 12*max_nodes bytes, and a
-  // 4-byte-aligned b64 LDS access measured 5x slower end-to-end.
-  int2* iwlds = (int2*)(((uintptr_t)(vlds + max_nodes) + 7) & ~7ull);
+  // instead of two b32 reads (the visit is LDS-instruction-bound)
+  int2* iwlds = (int2*)(vlds + max_nodes);    // [max_nodes][NNZ]
   KT* rows = (KT*)(iwlds + max_nodes * NNZ);  // [rows_per_iter][dpad]
 
   for (int64_t block_row0 = (int64_t)blockIdx.x * rows_per_iter; block_row0 < N;

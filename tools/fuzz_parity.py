@@ -137,7 +137,7 @@ def one_case(rs: np.random.RandomState, it: int) -> str:
         while (dpad_s % 4 != 2) if bf16 else (dpad_s % 2 != 1):
             dpad_s += 1
         sparse_route = (nnz <= 5
-                        and mn * (12 + nnz * 8) + 8 + 2 * 256 * dpad_s * elem
+                        and mn * (12 + nnz * 8) + 2 * 256 * dpad_s * elem
                         <= 150 * 1024)
         if d <= 8:
             D = 8

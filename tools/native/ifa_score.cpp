@@ -647,7 +647,7 @@ int main(int argc, char** argv) {
     bool try_sparse = nnz_seen <= 5;
     if (try_sparse) {
       PackedEIF p = pack_eif(m, 0, /*want_sparse=*/true);
-      size_t lds = (size_t)p.mn * (12 + p.nnz * 8) + 8 +
+      size_t lds = (size_t)p.mn * (12 + p.nnz * 8) +
                    (size_t)2 * 256 * dpad * elem;
       if (p.uniform && lds <= 150 * 1024) {
         T_report = p.T;
