@@ -733,7 +733,8 @@ def _device_forest(model, device, v4_key=None):
                 aos, values_t, _, max_depth = _eif_nodes_values_device(raw)
                 ncount = raw["ncount"]
                 extra["values"] = values_t
-                hidx_t, hw_t = raw["hidx"], raw["hw"]
+                extra["hidx"] = raw["hidx"]
+                extra["hw"] = raw["hw"]
             else:
                 packed, values, max_depth = _eif_nodes_values(forest)
                 aos = torch.from_numpy(packed).to(device)
@@ -741,15 +742,10 @@ def _device_forest(model, device, v4_key=None):
                     np.ascontiguousarray(forest.node_count, dtype=np.int32)
                 ).to(device)
                 extra["values"] = torch.from_numpy(values).to(device)
-                hidx_t = torch.from_numpy(
+                extra["hidx"] = torch.from_numpy(
                     np.ascontiguousarray(forest.hyper_idx)).to(device)
-                hw_t = torch.from_numpy(
+                extra["hw"] = torch.from_numpy(
                     np.ascontiguousarray(forest.hyper_w)).to(device)
-            # (idx, w-bits) interleaved int2 — one b64 LDS read per
-            # coordinate in the sparse kernel instead of two b32s
-            extra["iw"] = torch.stack(
-                [hidx_t.to(torch.int32), hw_t.view(torch.int32)],
-                dim=3).contiguous()
             extra["height"] = max_depth
         elif isinstance(v4_key, tuple) and v4_key[0] in ("eif_dense",
                                                          "eif_dense3"):
@@ -924,8 +920,8 @@ def score_extended_forest(model, X: torch.Tensor, finalize: bool = True) -> torc
             aos, ncount, extra = _device_forest(
                 model, X.device, v4_key="eif_sparse")
             return ext.score_extended_sparse_v2(
-                X.contiguous(), aos, extra["values"], extra["iw"],
-                ncount, extra["height"], c, finalize,
+                X.contiguous(), aos, extra["values"], extra["hidx"],
+                extra["hw"], ncount, extra["height"], c, finalize,
             )
     if d <= 128 and (nnz == d or nnz >= 6):
         if d <= 8:

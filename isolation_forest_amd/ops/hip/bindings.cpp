@@ -60,13 +60,13 @@ void launch_score_extended_forest(bool bf16, bool rows_lds, bool hyper_lds,
 
 void launch_score_extended_sparse_v2(bool bf16, int nnz, const void* X,
                                      const void* nodes, const float* values,
-                                     const void* iw, const int32_t* ncount,
-                                     float* out, int64_t N, int32_t d,
-                                     int32_t dpad, int32_t T,
-                                     int32_t max_nodes, int32_t height_limit,
-                                     float fT, float c_norm, int finalize,
-                                     size_t lds, int blocks,
-                                     hipStream_t stream);
+                                     const int32_t* hidx, const float* hw,
+                                     const int32_t* ncount, float* out,
+                                     int64_t N, int32_t d, int32_t dpad,
+                                     int32_t T, int32_t max_nodes,
+                                     int32_t height_limit, float fT,
+                                     float c_norm, int finalize, size_t lds,
+                                     int blocks, hipStream_t stream);
 
 void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
                                     const void* nodes, const float* values,
@@ -484,7 +484,7 @@ torch::Tensor score_extended_dense_v3(torch::Tensor X,
 torch::Tensor score_extended_sparse_v2(torch::Tensor X,
                                        torch::Tensor nodes_packed,
                                        torch::Tensor values,
-                                       torch::Tensor iw,
+                                       torch::Tensor hidx, torch::Tensor hw,
                                        torch::Tensor ncount,
                                        int64_t height_limit, double c_norm,
                                        bool finalize) {
@@ -494,16 +494,15 @@ torch::Tensor score_extended_sparse_v2(torch::Tensor X,
   CHECK_CONTIG(nodes_packed);
   CHECK_CUDA(values);
   CHECK_CONTIG(values);
-  CHECK_CUDA(iw);
-  CHECK_CONTIG(iw);
+  CHECK_CUDA(hidx);
+  CHECK_CONTIG(hidx);
+  CHECK_CUDA(hw);
+  CHECK_CONTIG(hw);
   CHECK_CUDA(ncount);
   check_x(X);
-  TORCH_CHECK(iw.dim() == 4 && iw.size(3) == 2 &&
-                  iw.scalar_type() == torch::kInt32,
-              "iw must be int32 [T, max_nodes, nnz, 2] (idx, w-bits)");
   int64_t N = X.size(0), d = X.size(1);
   int64_t T = nodes_packed.size(0), max_nodes = nodes_packed.size(1);
-  int64_t nnz = iw.size(2);
+  int64_t nnz = hidx.size(2);
   TORCH_CHECK(nnz >= 1 && nnz <= 5, "sparse v2 supports nnz <= 5");
   auto out = torch::empty({N}, X.options().dtype(torch::kFloat32));
   if (N == 0) return out;
@@ -517,13 +516,13 @@ torch::Tensor score_extended_sparse_v2(torch::Tensor X,
     while (dpad % 2 != 1) ++dpad;
   }
   size_t lds = (size_t)max_nodes * (12 + nnz * 8)
-               + (size_t)2 * 256 * dpad * elem;  // iw int2 == old idx+w bytes
+               + (size_t)2 * 256 * dpad * elem;
   TORCH_CHECK(lds <= 150 * 1024, "sparse v2 LDS overflow; use general path");
   int blocks = (int)std::min<int64_t>((N + 511) / 512, 8192);
   ifa::launch_score_extended_sparse_v2(
       bf16, (int)nnz, X.data_ptr(), nodes_packed.data_ptr<int32_t>(),
-      values.data_ptr<float>(), iw.data_ptr<int32_t>(),
-      ncount.data_ptr<int32_t>(),
+      values.data_ptr<float>(), hidx.data_ptr<int32_t>(),
+      hw.data_ptr<float>(), ncount.data_ptr<int32_t>(),
       out.data_ptr<float>(), N, (int32_t)d, (int32_t)dpad, (int32_t)T,
       (int32_t)max_nodes, (int32_t)height_limit, (float)T, (float)c_norm,
       finalize ? 1 : 0, lds, blocks, current_stream());

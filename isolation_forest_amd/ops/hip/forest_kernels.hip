@@ -1119,7 +1119,8 @@ __global__ void __launch_bounds__(256) score_extended_sparse_v2(
     const KT* __restrict__ X,           // raw bits [N][d]
     const int2* __restrict__ nodes,     // [T][max_nodes] {w0, offset/-inf}
     const float* __restrict__ values,   // [T][max_nodes] leaf value+depth
-    const int2* __restrict__ iw_g,      // [T][max_nodes][NNZ] {idx, w bits}
+    const int32_t* __restrict__ hidx_g, // [T][max_nodes][NNZ]
+    const float* __restrict__ hw_g,     // [T][max_nodes][NNZ]
     const int32_t* __restrict__ ncnt, float* __restrict__ out, int64_t N,
     int32_t d, int32_t dpad, int32_t T, int32_t max_nodes,
     int32_t height_limit, float fT, float c_norm, int32_t finalize) {
@@ -1128,10 +1129,9 @@ __global__ void __launch_bounds__(256) score_extended_sparse_v2(
 
   int2* tlds = (int2*)smem;                   // [max_nodes]
   float* vlds = (float*)(tlds + max_nodes);   // [max_nodes]
-  // (idx, weight) interleaved as int2: one ds_read_b64 per coordinate
-  // instead of two b32 reads (the visit is LDS-instruction-bound)
-  int2* iwlds = (int2*)(vlds + max_nodes);    // [max_nodes][NNZ]
-  KT* rows = (KT*)(iwlds + max_nodes * NNZ);  // [rows_per_iter][dpad]
+  int32_t* ilds = (int32_t*)(vlds + max_nodes);  // [max_nodes][NNZ]
+  float* wlds = (float*)(ilds + max_nodes * NNZ);  // [max_nodes][NNZ]
+  KT* rows = (KT*)(wlds + max_nodes * NNZ);   // [rows_per_iter][dpad]
 
   for (int64_t block_row0 = (int64_t)blockIdx.x * rows_per_iter; block_row0 < N;
        block_row0 += (int64_t)gridDim.x * rows_per_iter) {
@@ -1163,8 +1163,12 @@ __global__ void __launch_bounds__(256) score_extended_sparse_v2(
           tlds[i] = ss[i];
           vlds[i] = vs[i];
         }
-        const int2* iws = iw_g + (int64_t)t * max_nodes * NNZ;
-        for (int g = tid; g < nc * NNZ; g += 256) iwlds[g] = iws[g];
+        const int32_t* is = hidx_g + (int64_t)t * max_nodes * NNZ;
+        const float* ws = hw_g + (int64_t)t * max_nodes * NNZ;
+        for (int g = tid; g < nc * NNZ; g += 256) {
+          ilds[g] = is[g];
+          wlds[g] = ws[g];
+        }
       }
       __syncthreads();
 
@@ -1176,25 +1180,28 @@ __global__ void __launch_bounds__(256) score_extended_sparse_v2(
         int2 nd[RPT];
 #pragma unroll
         for (int r = 0; r < RPT; ++r) nd[r] = tlds[cur[r]];
-        int2 ciw[RPT][NNZ];
+        int32_t ci[RPT][NNZ];
 #pragma unroll
         for (int r = 0; r < RPT; ++r)
 #pragma unroll
-          for (int j = 0; j < NNZ; ++j) ciw[r][j] = iwlds[cur[r] * NNZ + j];
+          for (int j = 0; j < NNZ; ++j) ci[r][j] = ilds[cur[r] * NNZ + j];
+        float cw[RPT][NNZ];
+#pragma unroll
+        for (int r = 0; r < RPT; ++r)
+#pragma unroll
+          for (int j = 0; j < NNZ; ++j) cw[r][j] = wlds[cur[r] * NNZ + j];
         float xv[RPT][NNZ];
 #pragma unroll
         for (int r = 0; r < RPT; ++r)
 #pragma unroll
           for (int j = 0; j < NNZ; ++j)
-            xv[r][j] = load_row_f32<KT>(rb[r], ciw[r][j].x);
+            xv[r][j] = load_row_f32<KT>(rb[r], ci[r][j]);
 #pragma unroll
         for (int r = 0; r < RPT; ++r) {
           float dot = 0.f;
 #pragma unroll
           for (int j = 0; j < NNZ; ++j)  // oracle j-order
-            dot = __fadd_rn(dot,
-                            __fmul_rn(__int_as_float(ciw[r][j].y),
-                                      xv[r][j]));
+            dot = __fadd_rn(dot, __fmul_rn(cw[r][j], xv[r][j]));
           const int right = pn_right(nd[r].x);
           cur[r] = (dot < __int_as_float(nd[r].y)) ? cur[r] + 1 : right;
         }
@@ -1631,20 +1638,19 @@ void launch_score_extended_dense_v3(int D, bool rpt2, const void* X,
 
 void launch_score_extended_sparse_v2(bool bf16, int nnz, const void* X,
                                      const void* nodes, const float* values,
-                                     const void* iw, const int32_t* ncount,
-                                     float* out, int64_t N, int32_t d,
-                                     int32_t dpad, int32_t T,
-                                     int32_t max_nodes, int32_t height_limit,
-                                     float fT, float c_norm, int finalize,
-                                     size_t lds, int blocks,
-                                     hipStream_t stream) {
+                                     const int32_t* hidx, const float* hw,
+                                     const int32_t* ncount, float* out,
+                                     int64_t N, int32_t d, int32_t dpad,
+                                     int32_t T, int32_t max_nodes,
+                                     int32_t height_limit, float fT,
+                                     float c_norm, int finalize, size_t lds,
+                                     int blocks, hipStream_t stream) {
 #define LSS2(KT, NZ)                                                          \
   do {                                                                        \
     raise_lds((const void*)score_extended_sparse_v2<KT, NZ, 2>, lds);         \
     hipLaunchKernelGGL((score_extended_sparse_v2<KT, NZ, 2>), dim3(blocks),   \
                        dim3(256), lds, stream, (const KT*)X,                  \
-                       (const int2*)nodes, values, (const int2*)iw,          \
-                       ncount, out, N,                                        \
+                       (const int2*)nodes, values, hidx, hw, ncount, out, N,  \
                        d, dpad, T, max_nodes, height_limit, fT, c_norm,       \
                        finalize);                                             \
   } while (0)

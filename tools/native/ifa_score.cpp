@@ -59,13 +59,13 @@ void launch_score_extended_dense_v2(bool bf16, int D, const void* X,
                                     int blocks, hipStream_t stream);
 void launch_score_extended_sparse_v2(bool bf16, int nnz, const void* X,
                                      const void* nodes, const float* values,
-                                     const void* iw, const int32_t* ncount,
-                                     float* out, int64_t N, int32_t d,
-                                     int32_t dpad, int32_t T,
-                                     int32_t max_nodes, int32_t height_limit,
-                                     float fT, float c_norm, int finalize,
-                                     size_t lds, int blocks,
-                                     hipStream_t stream);
+                                     const int32_t* hidx, const float* hw,
+                                     const int32_t* ncount, float* out,
+                                     int64_t N, int32_t d, int32_t dpad,
+                                     int32_t T, int32_t max_nodes,
+                                     int32_t height_limit, float fT,
+                                     float c_norm, int finalize, size_t lds,
+                                     int blocks, hipStream_t stream);
 }
 
 #define HIP_CHECK(x)                                                         \
@@ -651,21 +651,19 @@ int main(int argc, char** argv) {
                    (size_t)2 * 256 * dpad * elem;
       if (p.uniform && lds <= 150 * 1024) {
         T_report = p.T;
-        void *dNodes, *dVals, *dHidx, *dNcount;
+        void *dNodes, *dVals, *dHidx, *dHw, *dNcount;
         HIP_CHECK(hipMalloc(&dNodes, p.nodes.size() * 4));
         HIP_CHECK(hipMalloc(&dVals, p.values.size() * 4));
-        std::vector<int32_t> iw(p.hidx.size() * 2);
-        for (size_t i = 0; i < p.hidx.size(); ++i) {
-          iw[2 * i] = p.hidx[i];
-          memcpy(&iw[2 * i + 1], &p.hw_sparse[i], 4);
-        }
-        HIP_CHECK(hipMalloc(&dHidx, iw.size() * 4));
+        HIP_CHECK(hipMalloc(&dHidx, p.hidx.size() * 4));
+        HIP_CHECK(hipMalloc(&dHw, p.hw_sparse.size() * 4));
         HIP_CHECK(hipMalloc(&dNcount, p.ncount.size() * 4));
         HIP_CHECK(hipMemcpy(dNodes, p.nodes.data(), p.nodes.size() * 4,
                             hipMemcpyHostToDevice));
         HIP_CHECK(hipMemcpy(dVals, p.values.data(), p.values.size() * 4,
                             hipMemcpyHostToDevice));
-        HIP_CHECK(hipMemcpy(dHidx, iw.data(), iw.size() * 4,
+        HIP_CHECK(hipMemcpy(dHidx, p.hidx.data(), p.hidx.size() * 4,
+                            hipMemcpyHostToDevice));
+        HIP_CHECK(hipMemcpy(dHw, p.hw_sparse.data(), p.hw_sparse.size() * 4,
                             hipMemcpyHostToDevice));
         HIP_CHECK(hipMemcpy(dNcount, p.ncount.data(), p.ncount.size() * 4,
                             hipMemcpyHostToDevice));
@@ -673,7 +671,8 @@ int main(int argc, char** argv) {
         if (bf16) { while (dpad_s % 4 != 2) ++dpad_s; }
         else { while (dpad_s % 2 != 1) ++dpad_s; }
         ifa::launch_score_extended_sparse_v2(
-            bf16, p.nnz, dX, dNodes, (const float*)dVals, dHidx,
+            bf16, p.nnz, dX, dNodes, (const float*)dVals,
+            (const int32_t*)dHidx, (const float*)dHw,
             (const int32_t*)dNcount, (float*)dOut, N, d, (int32_t)dpad_s,
             p.T, p.mn, p.max_depth, (float)p.T, c_norm, 1, lds, blocks, 0);
         goto launched;
