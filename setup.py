@@ -48,6 +48,7 @@ setup(
         "isolation_forest_amd.persist",
         "isolation_forest_amd.utils",
         "isolation_forest_amd.onnx",
+        "isolation_forest_amd.serving",
     ],
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
