@@ -12,15 +12,16 @@ def main():
                     help="cuda:N or cpu (default: cuda:0 when available)")
     ap.add_argument("--host", default="127.0.0.1")
     ap.add_argument("--port", type=int, default=8080)
-    ap.add_argument("--workers", type=int, default=1)
     args = ap.parse_args()
 
     import uvicorn
 
     from .app import create_app
 
+    # single worker by design: one model copy per process; scale out with
+    # multiple processes behind a load balancer (one per GPU)
     uvicorn.run(create_app(args.model, args.device),
-                host=args.host, port=args.port, workers=args.workers)
+                host=args.host, port=args.port)
 
 
 if __name__ == "__main__":
