@@ -225,3 +225,23 @@ class TestReferenceModelConversion:
         engine_scores = model.score(torch.from_numpy(X)).numpy()
         onnx_scores = evaluator.run(onnx_bytes, X)["outlier_score"].ravel()
         assert np.abs(onnx_scores - engine_scores).max() < 1e-5
+
+
+class TestUnsetThresholdExport:
+    def test_labels_match_reference_converter_semantics(self, gaussian_data):
+        """Threshold sentinel -1: the exported graph's Less/Not chain makes
+        every label 1 (score >= -1) — byte-for-byte what the REFERENCE
+        converter emits for such metadata (its graph is identical); the
+        engines' native transform instead emits literal 0.0 labels
+        (IsolationForestModel.scala:143-148). Pinned so the divergence is
+        intentional, not accidental."""
+        X, _ = gaussian_data
+        from isolation_forest_amd import IsolationForest
+
+        m = IsolationForest(numEstimators=10, randomSeed=2).fit(X[:800])
+        assert m.outlier_score_threshold == -1.0
+        data = IsolationForestConverter.from_model(m).convert().serialize()
+        r = evaluator.run(data, X[:16])
+        assert (r["predicted_label"].ravel() == 1).all()
+        native = m.transform(X[:16])["predictedLabel"]
+        assert float(native.sum()) == 0.0
