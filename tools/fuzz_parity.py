@@ -177,15 +177,21 @@ def one_case(rs: np.random.RandomState, it: int) -> str:
             # the flip FRACTION arbitrarily (every copy of a knife-edge
             # row flips together) but never the margin; an indexing bug
             # flips rows at large margins and fails here
+            # duplicate-heavy data can legitimately flip MOST rows: a
+            # pure-duplicate segment's offset is EXACTLY the class's own
+            # projection (every intercept has lo == hi), so the whole
+            # class sits on an exact f64 tie that f32 rounding resolves
+            # either way. No fraction ceiling therefore — instead a
+            # STRIDED sample across all flipped rows must each be
+            # individually margin-explained (a systematic inversion flips
+            # rows at large margins and fails).
             oracle_f = f_b if (bf16 and dense_route) else cpu_f
-            for r in bad[:80]:
+            step_ = max(1, len(bad) // 80)
+            for r in bad[::step_][:80]:
                 if not _knife_edge_explained(oracle_f, X[r]):
                     return (f"UNEXPLAINED dense mismatch row {int(r)} "
                             f"(diff {float(diff[r]):.5f}, "
                             f"{frac:.4f} rows off): {desc}")
-            if frac > 0.5:  # sanity ceiling even when sampled-explained
-                return (f"SCORE MISMATCH EIF dense (>{frac:.4f} rows off): "
-                        f"{desc}")
         else:  # sparse v2 or general strict-order kernel: bitwise
             if not np.array_equal(gpu_ps.cpu().numpy().view(np.int32),
                                   cpu_ps.view(np.int32)):
